@@ -1,0 +1,43 @@
+// Python bindings for the code_intelligence_amd gfx950 kernels.
+#include <torch/extension.h>
+#include <vector>
+
+namespace ci {
+void lstm_seq_forward_lib(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                          at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
+                          at::Tensor cs, at::Tensor gates);
+void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
+                            at::Tensor cs, at::Tensor gates);
+void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
+                       at::Tensor gates, at::Tensor hs, at::Tensor cs,
+                       at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
+                       at::Tensor dh0, at::Tensor dc0);
+at::Tensor concat_pool(at::Tensor hidden, at::Tensor lengths);
+void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                 at::Tensor tgt);
+void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                at::Tensor scale);
+void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                 std::vector<at::Tensor> masters, std::vector<at::Tensor> eas_,
+                 std::vector<at::Tensor> eass, double lr, double b1, double b2,
+                 double eps, double wd, double bc1, double bc2);
+at::Tensor emb_gather(at::Tensor weight, at::Tensor ids, at::Tensor rowmask);
+at::Tensor emb_scatter(at::Tensor gout, at::Tensor ids, at::Tensor rowmask,
+                       long V, long pad_idx);
+}  // namespace ci
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("lstm_seq_forward_lib", &ci::lstm_seq_forward_lib,
+        "LSTM sequence forward (hipBLASLt GEMM + pointwise cell kernel)");
+  m.def("lstm_seq_forward_fused", &ci::lstm_seq_forward_fused,
+        "LSTM sequence forward (fused MFMA cell kernel)");
+  m.def("lstm_seq_backward", &ci::lstm_seq_backward,
+        "LSTM sequence backward (pointwise + per-step GEMM)");
+  m.def("concat_pool", &ci::concat_pool, "masked mean/max/last concat pool");
+  m.def("ce_rowstats", &ci::ce_rowstats, "CE row logsumexp + target logit");
+  m.def("ce_dlogits", &ci::ce_dlogits, "in-place (softmax-onehot)*scale");
+  m.def("fused_adamw", &ci::fused_adamw, "fused AdamW step");
+  m.def("emb_gather", &ci::emb_gather, "embedding gather with row dropout");
+  m.def("emb_scatter", &ci::emb_scatter, "embedding grad scatter-add");
+}

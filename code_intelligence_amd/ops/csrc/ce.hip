@@ -1,0 +1,90 @@
+// K6 epilogues: fused tied-decoder softmax+CE row reductions.
+// The (chunked) logits GEMM runs on hipBLASLt; these kernels do the
+// per-row max/logsumexp pass and the in-place dlogits transform so full
+// logits never round-trip more than once (SURVEY.md §7 "hard parts": K6).
+#include "common.h"
+
+namespace ci {
+
+// one block per row: lse[i] = log(sum(exp(x - max))) + max ; tgt[i] = x[target]
+template <typename T, int THREADS>
+__global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride,
+                                   const long* __restrict__ targets,
+                                   float* __restrict__ lse,
+                                   float* __restrict__ tgt, int V) {
+  const int row = blockIdx.x;
+  const T* x = logits + (long)row * row_stride;
+  __shared__ float red[THREADS / kWave];
+  // pass 1: max
+  float mx = -3.4e38f;
+  for (int v = threadIdx.x; v < V; v += THREADS) mx = fmaxf(mx, ld(x + v));
+  #pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1)
+    mx = fmaxf(mx, __shfl_down(mx, off));
+  if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = mx;
+  __syncthreads();
+  if (threadIdx.x < THREADS / kWave) mx = red[threadIdx.x];
+  #pragma unroll
+  for (int off = THREADS / kWave / 2; off > 0; off >>= 1)
+    mx = fmaxf(mx, __shfl_down(mx, off));
+  mx = __shfl(mx, 0);
+  __syncthreads();
+  // pass 2: sum exp
+  float s = 0.f;
+  for (int v = threadIdx.x; v < V; v += THREADS) s += __expf(ld(x + v) - mx);
+  #pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
+  if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float tot = 0.f;
+    for (int w = 0; w < THREADS / kWave; ++w) tot += red[w];
+    lse[row] = mx + __logf(tot);
+    tgt[row] = ld(x + targets[row]);
+  }
+}
+
+void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                 at::Tensor tgt) {
+  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
+  const int N = logits.size(0), V = logits.size(1);
+  constexpr int THREADS = 256;
+  CI_DISPATCH_FB(logits.scalar_type(), "ce_rowstats", [&] {
+    hipLaunchKernelGGL((ce_rowstats_kernel<scalar_t, THREADS>), dim3(N),
+        dim3(THREADS), 0, stream(),
+        reinterpret_cast<const scalar_t*>(logits.data_ptr()), (long)V,
+        targets.data_ptr<long>(), lse.data_ptr<float>(), tgt.data_ptr<float>(), V);
+  });
+}
+
+// in-place: logits <- (exp(logits - lse) - onehot) * scale   (scale on device)
+template <typename T>
+__global__ void ce_dlogits_kernel(T* __restrict__ logits,
+                                  const long* __restrict__ targets,
+                                  const float* __restrict__ lse,
+                                  const float* __restrict__ scale,
+                                  int N, int V) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)N * V) return;
+  const int row = idx / V, col = idx % V;
+  float p = __expf(ld(logits + idx) - lse[row]);
+  if (col == (int)targets[row]) p -= 1.f;
+  st(logits + idx, p * scale[0]);
+}
+
+void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                at::Tensor scale) {
+  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
+  const int N = logits.size(0), V = logits.size(1);
+  const int threads = 256;
+  const long total = (long)N * V;
+  CI_DISPATCH_FB(logits.scalar_type(), "ce_dlogits", [&] {
+    hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t>),
+        dim3(ceil_div(total, threads)), dim3(threads), 0, stream(),
+        reinterpret_cast<scalar_t*>(logits.data_ptr()),
+        targets.data_ptr<long>(), lse.data_ptr<float>(),
+        scale.data_ptr<float>(), N, V);
+  });
+}
+
+}  // namespace ci

@@ -1,0 +1,64 @@
+// Common helpers for the code_intelligence_amd gfx950 HIP kernels.
+// Target: AMD Instinct MI355X (CDNA4, wave64, 256 CUs / 8 XCDs).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define CI_CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be on device")
+#define CI_CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+namespace ci {
+
+constexpr int kWave = 64;  // CDNA wavefront
+
+static inline hipStream_t stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---- scalar conversion helpers ------------------------------------------
+template <typename T> struct Cvt;
+template <> struct Cvt<float> {
+  static __device__ __forceinline__ float load(const float* p) { return *p; }
+  static __device__ __forceinline__ void store(float* p, float v) { *p = v; }
+};
+template <> struct Cvt<__hip_bfloat16> {
+  static __device__ __forceinline__ float load(const __hip_bfloat16* p) {
+    return __bfloat162float(*p);
+  }
+  static __device__ __forceinline__ void store(__hip_bfloat16* p, float v) {
+    *p = __float2bfloat16(v);
+  }
+};
+
+template <typename T>
+static __device__ __forceinline__ float ld(const T* p) { return Cvt<T>::load(p); }
+template <typename T>
+static __device__ __forceinline__ void st(T* p, float v) { Cvt<T>::store(p, v); }
+
+static __device__ __forceinline__ float sigmoidf_(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+// dispatch on an ATen dtype (float32 / bfloat16)
+#define CI_DISPATCH_FB(DTYPE, NAME, ...)                                   \
+  [&] {                                                                    \
+    switch (DTYPE) {                                                       \
+      case at::ScalarType::Float: {                                        \
+        using scalar_t = float;                                            \
+        return __VA_ARGS__();                                              \
+      }                                                                    \
+      case at::ScalarType::BFloat16: {                                     \
+        using scalar_t = __hip_bfloat16;                                   \
+        return __VA_ARGS__();                                              \
+      }                                                                    \
+      default:                                                             \
+        TORCH_CHECK(false, NAME ": unsupported dtype ", DTYPE);            \
+    }                                                                      \
+  }()
+
+static inline int ceil_div(long a, long b) { return (int)((a + b - 1) / b); }
+
+}  // namespace ci

@@ -1,0 +1,155 @@
+// K2 (lib mode): LSTM cell pointwise kernels + C++ sequence drivers.
+// The recurrent GEMM runs through hipBLASLt (at::mm) per timestep; the gate
+// activation + c/h update (and their backward) are fused HIP kernels.
+// Reference semantics: fastai AWD_LSTM nn.LSTM cell, gate order i,f,g,o
+// (/root/reference/Issue_Embeddings/train.py:88-92).
+#include "common.h"
+
+namespace ci {
+
+// one thread per (b, j): gates = xp + rec (+bias); c' = f*c + i*g; h' = o*tanh(c')
+template <typename T>
+__global__ void lstm_cell_fwd(
+    const T* __restrict__ xp, long xp_rs,      // (B,4H) view, row stride xp_rs
+    const T* __restrict__ rec, long rec_rs,    // (B,4H) recurrent GEMM out
+    const float* __restrict__ bias,            // (4H) b_ih + b_hh
+    const float* __restrict__ c_prev, long cp_rs,
+    T* __restrict__ h_out, long h_rs,
+    float* __restrict__ c_out, long c_rs,
+    T* __restrict__ gates_out, long g_rs,      // post-activation i,f,g,o
+    int B, int H) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * H) return;
+  const int b = idx / H, j = idx % H;
+  const long xo = (long)b * xp_rs + j;
+  const long ro = (long)b * rec_rs + j;
+  float gi = ld(xp + xo) + ld(rec + ro) + bias[j];
+  float gf = ld(xp + xo + H) + ld(rec + ro + H) + bias[j + H];
+  float gg = ld(xp + xo + 2 * H) + ld(rec + ro + 2 * H) + bias[j + 2 * H];
+  float go = ld(xp + xo + 3 * H) + ld(rec + ro + 3 * H) + bias[j + 3 * H];
+  gi = sigmoidf_(gi); gf = sigmoidf_(gf); gg = tanhf(gg); go = sigmoidf_(go);
+  const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
+  const float h = go * tanhf(c);
+  st(h_out + (long)b * h_rs + j, h);
+  c_out[(long)b * c_rs + j] = c;
+  const long gout = (long)b * g_rs + j;
+  st(gates_out + gout, gi);
+  st(gates_out + gout + H, gf);
+  st(gates_out + gout + 2 * H, gg);
+  st(gates_out + gout + 3 * H, go);
+}
+
+// backward pointwise: consumes dh_ext (from upstream) + dh_rec (from t+1 GEMM),
+// running dc (fp32 buffer, in/out), saved post-act gates, c_{t-1}, c_t.
+template <typename T>
+__global__ void lstm_cell_bwd(
+    const T* __restrict__ dh_ext, long dhe_rs,
+    const T* __restrict__ dh_rec, long dhr_rs,
+    float* __restrict__ dc_buf, long dc_rs,    // in: dc_t ; out: dc_{t-1}
+    const T* __restrict__ gates, long g_rs,
+    const float* __restrict__ c_prev, long cp_rs,
+    const float* __restrict__ c_t, long ct_rs,
+    T* __restrict__ dgates, long dg_rs,
+    int B, int H) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * H) return;
+  const int b = idx / H, j = idx % H;
+  const long go = (long)b * g_rs + j;
+  const float gi = ld(gates + go);
+  const float gf = ld(gates + go + H);
+  const float gg = ld(gates + go + 2 * H);
+  const float gout = ld(gates + go + 3 * H);
+  float dh = ld(dh_ext + (long)b * dhe_rs + j) + ld(dh_rec + (long)b * dhr_rs + j);
+  const float ct = c_t[(long)b * ct_rs + j];
+  const float tct = tanhf(ct);
+  const float do_ = dh * tct;
+  float dct = dc_buf[(long)b * dc_rs + j] + dh * gout * (1.f - tct * tct);
+  const float di = dct * gg;
+  const float df = dct * c_prev[(long)b * cp_rs + j];
+  const float dg = dct * gi;
+  dc_buf[(long)b * dc_rs + j] = dct * gf;  // dc_{t-1}
+  const long dgo = (long)b * dg_rs + j;
+  st(dgates + dgo, di * gi * (1.f - gi));
+  st(dgates + dgo + H, df * gf * (1.f - gf));
+  st(dgates + dgo + 2 * H, dg * (1.f - gg * gg));
+  st(dgates + dgo + 3 * H, do_ * gout * (1.f - gout));
+}
+
+template <typename ST>
+static void launch_fwd_step(const at::Tensor& xp, const at::Tensor& bias,
+                            const at::Tensor& rec, const at::Tensor& c_prev,
+                            long cp_off, long cp_rs, at::Tensor& hs,
+                            at::Tensor& cs, at::Tensor& gates, int t, int B,
+                            int T, int H) {
+  const int threads = 256;
+  const int blocks = ceil_div((long)B * H, threads);
+  hipLaunchKernelGGL((lstm_cell_fwd<ST>), dim3(blocks), dim3(threads), 0, stream(),
+      reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+      reinterpret_cast<const ST*>(rec.data_ptr()), (long)4 * H,
+      bias.data_ptr<float>(),
+      c_prev.data_ptr<float>() + cp_off, cp_rs,
+      reinterpret_cast<ST*>(hs.data_ptr()) + (long)t * H, (long)T * H,
+      cs.data_ptr<float>() + (long)t * H, (long)T * H,
+      reinterpret_cast<ST*>(gates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+      B, H);
+}
+
+// hs,cs,gates are (B,T,·) preallocated; xp (B,T,4H); h0 (B,H); c0 fp32 (B,H).
+void lstm_seq_forward_lib(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                          at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
+                          at::Tensor cs, at::Tensor gates) {
+  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
+  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates);
+  const int B = xp.size(0), T = xp.size(1);
+  const int H = w_hh.size(1);
+  auto w_hh_t = w_hh.t();
+  auto rec = at::empty({B, 4 * H}, xp.options());
+  CI_DISPATCH_FB(xp.scalar_type(), "lstm_seq_forward_lib", [&] {
+    for (int t = 0; t < T; ++t) {
+      auto h_prev = (t == 0) ? h0 : hs.select(1, t - 1);
+      at::mm_out(rec, h_prev, w_hh_t);
+      if (t == 0) {
+        launch_fwd_step<scalar_t>(xp, bias, rec, c0, 0, H, hs, cs, gates, t, B, T, H);
+      } else {
+        launch_fwd_step<scalar_t>(xp, bias, rec, cs, (long)(t - 1) * H,
+                                  (long)T * H, hs, cs, gates, t, B, T, H);
+      }
+    }
+  });
+}
+
+// reverse loop; dh0/dc0 are (B,H) fp32 outputs.
+void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
+                       at::Tensor gates, at::Tensor hs, at::Tensor cs,
+                       at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
+                       at::Tensor dh0, at::Tensor dc0) {
+  CI_CHECK_CUDA(dhs); CI_CHECK_CONTIG(dhs); CI_CHECK_CONTIG(dgates);
+  const int B = dhs.size(0), T = dhs.size(1);
+  const int H = w_hh.size(1);
+  auto dc_buf = dcT.clone();                       // (B,H) fp32 running dc
+  auto dh_rec = dhT.contiguous();                  // (B,H) scalar running rec grad
+  const int threads = 256;
+  const int blocks = ceil_div((long)B * H, threads);
+  CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
+    for (int t = T - 1; t >= 0; --t) {
+      const float* cprev = (t == 0) ? c0.data_ptr<float>()
+                                    : cs.data_ptr<float>() + (long)(t - 1) * H;
+      const long cprs = (t == 0) ? H : (long)T * H;
+      hipLaunchKernelGGL((lstm_cell_bwd<scalar_t>), dim3(blocks), dim3(threads), 0, stream(),
+          reinterpret_cast<const scalar_t*>(dhs.data_ptr()) + (long)t * H, (long)T * H,
+          reinterpret_cast<const scalar_t*>(dh_rec.data_ptr()), (long)H,
+          dc_buf.data_ptr<float>(), (long)H,
+          reinterpret_cast<const scalar_t*>(gates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+          cprev, cprs,
+          cs.data_ptr<float>() + (long)t * H, (long)T * H,
+          reinterpret_cast<scalar_t*>(dgates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+          B, H);
+      // dh_{t-1} recurrent contribution: dgates_t @ w_hh
+      at::mm_out(dh_rec, dgates.select(1, t), w_hh);
+    }
+  });
+  dh0.copy_(dh_rec);
+  dc0.copy_(dc_buf);
+}
+
+}  // namespace ci
