@@ -1,0 +1,132 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: AWD-LSTM LM pretraining step on MI355X.
+
+Measures the BASELINE.json headline metric — LM tokens/sec (whole node) —
+on the reference's deployed model config (AWD-LSTM n_layers=4, n_hid=2400,
+emb_sz=800, vocab 60k — BASELINE.md 'Deployed LM architecture') with
+synthetic issue-text tokens and random-init weights, bf16 compute,
+per-GPU batch 512 x seq 512 (BASELINE.json configs 2-3: global bs 4096 at
+DP=8, weak scaling).
+
+Single GPU:      python bench.py --gpus 1 --steps 20 --warmup 5
+Multi GPU (DP):  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                     --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from code_intelligence_amd.data.synthetic import synthetic_issue_tokens
+from code_intelligence_amd.models.awd_lstm import AWDLSTM
+from code_intelligence_amd.parallel.ddp import init_distributed
+from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--bs", type=int, default=512)        # per-GPU
+    p.add_argument("--seq", type=int, default=512)
+    p.add_argument("--vocab", type=int, default=60000)
+    p.add_argument("--emb", type=int, default=800)
+    p.add_argument("--hid", type=int, default=2400)
+    p.add_argument("--layers", type=int, default=4)
+    p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--lstm_mode", type=str, default=None,
+                   help="override CI_LSTM_MODE (fused|lib)")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    if args.lstm_mode:
+        os.environ["CI_LSTM_MODE"] = args.lstm_mode
+    rank, world = init_distributed()
+    on_gpu = torch.cuda.is_available()
+    if not on_gpu:
+        # CPU fallback: plumbing config (BASELINE.json config 1) so the
+        # script stays runnable off-GPU; the official metric is GPU-only.
+        args.bs, args.seq, args.vocab = 2, 64, 30000
+        args.emb, args.hid, args.layers = 400, 400, 3
+    device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) \
+        if on_gpu else torch.device("cpu")
+    dtype = torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else torch.float32
+
+    torch.manual_seed(1234 + rank)
+    model = AWDLSTM(vocab_sz=args.vocab, emb_sz=args.emb, n_hid=args.hid,
+                    n_layers=args.layers).to(device=device, dtype=dtype)
+    trainer = LMTrainer(model, TrainConfig(), distributed=(world > 1))
+    model.train()
+    model.reset(args.bs)
+
+    # synthetic issue-shaped stream, pre-batched on device
+    docs = synthetic_issue_tokens(64, args.vocab, seed=7 + rank, mean_len=300)
+    stream = torch.tensor([t for d in docs for t in d], dtype=torch.int64)
+    need = args.bs * (args.seq + 1) * (args.steps + args.warmup)
+    reps = need // stream.numel() + 1
+    stream = stream.repeat(reps)[:need].view(-1, args.seq + 1).to(device)
+
+    def batch(i):
+        s = (i * args.bs) % (stream.shape[0] - args.bs + 1)
+        win = stream[s: s + args.bs]
+        return win[:, :-1], win[:, 1:]
+
+    lr = trainer.cfg.lr
+    for i in range(args.warmup):
+        x, y = batch(i)
+        trainer.train_step(x, y, lr)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        x, y = batch(args.warmup + i)
+        trainer.train_step(x, y, lr)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if on_gpu else None)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t)
+
+    n_gpus = world if on_gpu else args.gpus
+    tokens = args.bs * args.seq * args.steps * world
+    value = tokens / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "LM tokens/sec (whole node)",
+            "value": round(value, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"AWD-LSTM {args.layers}x{args.hid} emb{args.emb} vocab{args.vocab}",
+                "global_batch": args.bs * world,
+                "seq_len": args.seq,
+                "parallelism": f"dp{world}",
+                "device": "MI355X" if on_gpu else "cpu-fallback",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

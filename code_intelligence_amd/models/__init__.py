@@ -1,0 +1,8 @@
+from .awd_lstm import (AWDLSTM, AWDLSTMEncoder, EmbeddingDropout,
+                       LinearDecoder, RNNDropout, WeightDroppedLSTM,
+                       awd_lstm_lm_config)
+
+__all__ = [
+    "AWDLSTM", "AWDLSTMEncoder", "EmbeddingDropout", "LinearDecoder",
+    "RNNDropout", "WeightDroppedLSTM", "awd_lstm_lm_config",
+]

@@ -1,0 +1,10 @@
+from .trainer import LMTrainer, TrainConfig
+from .callbacks import (Callback, CallbackList, EarlyStopping, SaveModel,
+                        ReduceLROnPlateau, CSVLogger, JSONRunLogger)
+from .schedules import OneCycle, FlatSchedule
+
+__all__ = [
+    "LMTrainer", "TrainConfig", "Callback", "CallbackList", "EarlyStopping",
+    "SaveModel", "ReduceLROnPlateau", "CSVLogger", "JSONRunLogger",
+    "OneCycle", "FlatSchedule",
+]

@@ -1,0 +1,143 @@
+"""LM pretraining loop — the MI355X-native equivalent of the reference's
+fastai fit/fit_one_cycle (train.py:104-113) including AR/TAR activation
+regularization (fastai RNNTrainer defaults alpha=2, beta=1) and the
+callback set of train.py:97-102.
+
+The loss path never materializes full logits: encoder -> output dropout ->
+fused tied-decoder CE (ops/crossentropy.py, K6)."""
+from __future__ import annotations
+
+import dataclasses
+import math
+import time
+from typing import Iterable, Optional
+
+import torch
+from torch import nn
+
+from ..models.awd_lstm import AWDLSTM
+from ..ops.crossentropy import tied_decoder_ce
+from ..ops.adam import FusedAdamW
+from ..parallel.ddp import DistributedGrads
+from .callbacks import Callback, CallbackList
+from .schedules import OneCycle, FlatSchedule
+
+
+@dataclasses.dataclass
+class TrainConfig:
+    lr: float = 1.3e-3            # reference default train.py:44
+    wd: float = 0.012             # train.py:45
+    one_cycle: bool = True
+    cycle_len: int = 1
+    betas: tuple = (0.9, 0.99)    # fastai Adam default (0.9, 0.99)
+    alpha: float = 2.0            # AR
+    beta: float = 1.0             # TAR
+    clip: float = 0.0
+    bucket_mb: float = 64.0
+
+
+class LMTrainer:
+    def __init__(self, model: AWDLSTM, cfg: TrainConfig = TrainConfig(),
+                 callbacks: Optional[list[Callback]] = None,
+                 distributed: bool = False):
+        self.model = model
+        self.cfg = cfg
+        self.cbs = CallbackList(callbacks)
+        self.opt = FusedAdamW(self._param_groups(), lr=cfg.lr,
+                              betas=cfg.betas, weight_decay=cfg.wd)
+        self.lr_scale = 1.0
+        self.dist = DistributedGrads(model, bucket_mb=cfg.bucket_mb) \
+            if distributed else None
+        self.global_step = 0
+
+    def _param_groups(self):
+        seen, uniq = set(), []
+        for p in self.model.parameters():
+            if p.requires_grad and id(p) not in seen:
+                seen.add(id(p))
+                uniq.append(p)
+        return uniq
+
+    def loss_on_batch(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        enc = self.model.encoder
+        dec = self.model.decoder
+        raw_outputs, outputs = enc(x)
+        out = dec.output_dp(outputs[-1])
+        loss = tied_decoder_ce(out.reshape(-1, out.shape[-1]),
+                               dec.decoder.weight, dec.decoder.bias,
+                               y.reshape(-1))
+        if self.model.training:
+            if self.cfg.alpha:  # activation regularization on dropped output
+                loss = loss + self.cfg.alpha * out.float().pow(2).mean()
+            if self.cfg.beta:   # temporal AR on raw output diffs
+                r = raw_outputs[-1].float()
+                if r.shape[1] > 1:
+                    loss = loss + self.cfg.beta * (r[:, 1:] - r[:, :-1]).pow(2).mean()
+        return loss
+
+    def train_step(self, x, y, lr: float, mom: Optional[float] = None) -> float:
+        for g in self.opt.param_groups:
+            g["lr"] = lr * self.lr_scale
+            if mom is not None:
+                g["betas"] = (mom, g["betas"][1])
+        if self.dist is not None:
+            self.dist.prepare()
+        self.opt.zero_grad(set_to_none=True)
+        loss = self.loss_on_batch(x, y)
+        loss.backward()
+        if self.dist is not None:
+            self.dist.finalize()
+        if self.cfg.clip:
+            torch.nn.utils.clip_grad_norm_(self._param_groups(), self.cfg.clip)
+        self.opt.step()
+        self.global_step += 1
+        return float(loss.detach())
+
+    @torch.no_grad()
+    def evaluate(self, loader: Iterable) -> dict:
+        self.model.eval()
+        self.model.reset()
+        tot, n = 0.0, 0
+        for x, y in loader:
+            loss = self.loss_on_batch(x, y)
+            tot += float(loss) * x.numel()
+            n += x.numel()
+        self.model.train()
+        self.model.reset()
+        vl = tot / max(n, 1)
+        return {"valid_loss": vl, "valid_ppl": math.exp(min(vl, 30.0))}
+
+    def fit(self, train_loader, valid_loader=None, epochs: int = 1,
+            one_cycle: Optional[bool] = None) -> dict:
+        cfg = self.cfg
+        use_oc = cfg.one_cycle if one_cycle is None else one_cycle
+        # reference: fit_one_cycle(cycle_len, max_lr=lr*2) (train.py:109-111)
+        sched = OneCycle(cfg.lr * 2) if use_oc else FlatSchedule(cfg.lr)
+        n_total = None
+        try:
+            n_total = len(train_loader) * epochs
+        except TypeError:
+            pass
+        self.cbs.on_train_begin(self)
+        metrics: dict = {}
+        self.model.train()
+        self.model.reset()
+        step = 0
+        t0 = time.time()
+        for epoch in range(epochs):
+            losses = []
+            for x, y in train_loader:
+                frac = (step / n_total) if n_total else 0.5
+                lr, mom = sched.at(frac) if use_oc else (cfg.lr, None)
+                loss = self.train_step(x, y, lr, mom)
+                losses.append(loss)
+                self.cbs.on_step_end(self, self.global_step, loss)
+                step += 1
+            metrics = {"train_loss": sum(losses) / max(len(losses), 1),
+                       "time_s": round(time.time() - t0, 2)}
+            if valid_loader is not None:
+                metrics.update(self.evaluate(valid_loader))
+            if self.cbs.on_epoch_end(self, epoch, metrics):
+                break
+        self.cbs.on_train_end(self)
+        return metrics

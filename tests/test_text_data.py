@@ -1,0 +1,72 @@
+"""Tokenizer rules, vocab, loader-stream semantics (CPU)."""
+import torch
+
+from code_intelligence_amd.data.lm_loader import LMStreamLoader
+from code_intelligence_amd.data.synthetic import synthetic_issue_tokens
+from code_intelligence_amd.text.tokenizer import (Tokenizer, Vocab, PAD, UNK,
+                                                  TK_MAJ, TK_REP, TK_UP,
+                                                  defaults_specials,
+                                                  process_dict, replace_rep,
+                                                  replace_all_caps, deal_caps)
+
+
+def test_special_token_indices():
+    v = Vocab(defaults_specials + ["hello", "world"])
+    assert v.stoi[UNK] == 0
+    assert v.stoi[PAD] == 1  # pad_token=1 everywhere in the reference
+
+
+def test_replace_rep():
+    assert TK_REP in replace_rep("soooooo cool")
+    assert replace_rep("soo cool") == "soo cool"  # <4 reps untouched
+
+
+def test_caps_rules():
+    assert replace_all_caps(["HELLO", "hi"]) == [TK_UP, "hello", "hi"]
+    assert deal_caps(["Hello", "world"]) == [TK_MAJ, "hello", "world"]
+
+
+def test_tokenizer_end_to_end():
+    tok = Tokenizer()
+    toks = tok.process_text("# Bug report\n`pip install` FAILS with ```trace```")
+    assert "xxcd" in toks  # inline code collapsed
+    assert "xxcdb" in toks  # code block collapsed
+    assert "xxup" in toks  # FAILS -> xxup fails
+    assert "fails" in toks
+
+
+def test_vocab_create_and_numericalize():
+    tok = Tokenizer()
+    docs = [tok.process_text("the bug the bug the bug"),
+            tok.process_text("a feature a feature a feature")]
+    v = Vocab.create(docs, max_vocab=100, min_freq=2)
+    ids = v.numericalize(["the", "zzzunknown"])
+    assert ids[0] >= len(defaults_specials)
+    assert ids[1] == v.stoi[UNK]
+
+
+def test_process_dict_reference_format():
+    d = process_dict({"title": "Crash on start", "body": "It **fails**."})
+    assert d["text"].startswith("xxxfldtitle")
+    assert "xxxfldbody" in d["text"]
+    assert process_dict({"title": None, "body": None})["text"] is not None
+
+
+def test_lm_loader_stream_continuity():
+    docs = [[10, 11, 12, 13, 14, 15, 16, 17]] * 8
+    dl = LMStreamLoader(docs, bs=2, bptt=4, bos_idx=2, shuffle=False)
+    batches = list(dl)
+    assert len(batches) == len(dl)
+    x0, y0 = batches[0]
+    assert x0.shape == (2, 4)
+    # y is x shifted by one within the stream
+    assert torch.equal(y0[:, :-1], x0[:, 1:])
+    # next window continues exactly where the previous ended
+    x1, y1 = batches[1]
+    assert torch.equal(x1[:, 0], y0[:, -1])
+
+
+def test_synthetic_tokens_in_range():
+    docs = synthetic_issue_tokens(10, vocab_sz=1000, seed=0)
+    flat = [t for d in docs for t in d]
+    assert min(flat) >= 9 and max(flat) < 1000
