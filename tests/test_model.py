@@ -112,5 +112,5 @@ def test_loss_decreases_tiny_train():
     x = torch.randint(9, 64, (4, 16))
     y = torch.roll(x, -1, dims=1)
     m.train()
-    losses = [tr.train_step(x, y, 5e-3) for _ in range(80)]
-    assert losses[-1] < losses[0] * 0.7, losses[::10]
+    losses = [tr.train_step(x, y, 1e-2) for _ in range(400)]
+    assert losses[-1] < 1.0, losses[::50]  # memorizes 4 fixed sequences
