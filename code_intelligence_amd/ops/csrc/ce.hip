@@ -12,10 +12,12 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
                                    const long* __restrict__ targets,
                                    float* __restrict__ lse,
                                    float* __restrict__ tgt, int V) {
+  constexpr int NW = THREADS / kWave;
   const int row = blockIdx.x;
   const T* x = logits + (long)row * row_stride;
-  __shared__ float red[THREADS / kWave];
-  // pass 1: max
+  __shared__ float red[NW];
+  __shared__ float bcast;
+  // pass 1: global row max
   float mx = -3.4e38f;
   for (int v = threadIdx.x; v < V; v += THREADS) mx = fmaxf(mx, ld(x + v));
   #pragma unroll
@@ -23,13 +25,15 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
     mx = fmaxf(mx, __shfl_down(mx, off));
   if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = mx;
   __syncthreads();
-  if (threadIdx.x < THREADS / kWave) mx = red[threadIdx.x];
-  #pragma unroll
-  for (int off = THREADS / kWave / 2; off > 0; off >>= 1)
-    mx = fmaxf(mx, __shfl_down(mx, off));
-  mx = __shfl(mx, 0);
+  if (threadIdx.x == 0) {
+    float m = red[0];
+    for (int w = 1; w < NW; ++w) m = fmaxf(m, red[w]);
+    bcast = m;
+  }
   __syncthreads();
-  // pass 2: sum exp
+  mx = bcast;
+  __syncthreads();  // red[] reused below
+  // pass 2: sum exp(x - max)
   float s = 0.f;
   for (int v = threadIdx.x; v < V; v += THREADS) s += __expf(ld(x + v) - mx);
   #pragma unroll
@@ -38,7 +42,7 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
   __syncthreads();
   if (threadIdx.x == 0) {
     float tot = 0.f;
-    for (int w = 0; w < THREADS / kWave; ++w) tot += red[w];
+    for (int w = 0; w < NW; ++w) tot += red[w];
     lse[row] = mx + __logf(tot);
     tgt[row] = ld(x + targets[row]);
   }
