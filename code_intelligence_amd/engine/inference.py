@@ -1,0 +1,165 @@
+"""InferenceWrapper — the embedding inference engine (serve hot path).
+
+API-compatible re-creation of py/code_intelligence/inference.py:25-263:
+``process_dict``, ``get_pooled_features`` (-> (1, 3*emb_sz) = 2400-d at the
+deployed config), ``df_to_embedding`` (sort-by-length batching, padding,
+OOM backoff halving bs — inference.py:138-229), ``batch_seq_pool``
+(length-masked pooling, inference.py:232-263 — here the K5 HIP kernel).
+
+MI355X design: the encoder runs the fused CDNA4 LSTM kernels; fixed-length
+buckets can be hipGraph-captured (engine/graph_exec.py) to kill the
+per-timestep launch overhead that dominates at serve batch sizes.
+"""
+from __future__ import annotations
+
+import json
+import math
+from pathlib import Path
+from typing import List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from ..models.awd_lstm import AWDLSTM, AWDLSTMEncoder
+from ..ops.pool import concat_pool
+from ..text.tokenizer import Tokenizer, Vocab, process_dict as _process_dict
+
+
+class InferenceWrapper:
+    """Loads model artifacts and turns (title, body) into pooled embeddings."""
+
+    def __init__(self, model_path: Optional[str] = None,
+                 model_file_name: Optional[str] = None,
+                 encoder: Optional[AWDLSTMEncoder] = None,
+                 vocab: Optional[Vocab] = None,
+                 device: Optional[str] = None,
+                 dtype: Optional[torch.dtype] = None,
+                 use_graphs: bool = False):
+        self.device = torch.device(device) if device else (
+            torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu"))
+        self.dtype = dtype or (torch.bfloat16 if self.device.type == "cuda"
+                               else torch.float32)
+        self.tokenizer = Tokenizer()
+        if encoder is not None:
+            assert vocab is not None
+            self.encoder, self.vocab = encoder, vocab
+        else:
+            root = Path(model_path)
+            if model_file_name:
+                root = root / model_file_name
+            cfg = json.loads((root / "config.json").read_text())
+            self.vocab = Vocab.load(root / "vocab.json")
+            model = AWDLSTM(vocab_sz=len(self.vocab), emb_sz=cfg["emb_sz"],
+                            n_hid=cfg["n_hid"], n_layers=cfg["n_layers"])
+            enc_file = root / cfg.get("encoder_file", "encoder.pth")
+            model.load_encoder(enc_file)
+            self.encoder = model.encoder
+        self.encoder = self.encoder.to(device=self.device, dtype=self.dtype)
+        self.encoder.eval()
+        self.emb_sz = self.encoder.emb_sz
+        self.pad_idx = self.encoder.pad_token
+        self._graphs = {}
+        self.use_graphs = use_graphs and self.device.type == "cuda"
+
+    # --- reference-parity helpers -----------------------------------------
+    def process_dict(self, data: dict) -> dict:
+        return _process_dict(data, self.tokenizer)
+
+    def numericalize(self, text: str) -> List[int]:
+        return self.vocab.numericalize(self.tokenizer.process_text(text))
+
+    @staticmethod
+    def _bucket(n: int, buckets=(1, 2, 4, 8, 16, 32, 64, 128, 256, 512, 1024, 2048)) -> int:
+        for b in buckets:
+            if n <= b:
+                return b
+        return n
+
+    @torch.no_grad()
+    def _encode_batch(self, ids: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
+        """ids: (B, T) padded; returns (B, 3*emb_sz) pooled fp32."""
+        B, T = ids.shape
+        if self.use_graphs:
+            # pad (B, T) up to fixed buckets so each shape is captured once
+            Bb, Tb = self._bucket(B), self._bucket(T)
+            if (Bb, Tb) != (B, T):
+                padded = torch.full((Bb, Tb), self.pad_idx, dtype=ids.dtype,
+                                    device=ids.device)
+                padded[:B, :T] = ids
+                ids = padded
+            from .graph_exec import GraphedEncoder
+            key = (Bb, Tb)
+            if key not in self._graphs:
+                self._graphs[key] = GraphedEncoder(self.encoder, Bb, Tb, self.device)
+            hidden = self._graphs[key].run(ids)[:B]
+        else:
+            self.encoder.reset(B)
+            _, outputs = self.encoder(ids)
+            hidden = outputs[-1]
+        return concat_pool(hidden, lengths).float()
+
+    @torch.no_grad()
+    def get_pooled_features(self, text: str) -> torch.Tensor:
+        """(1, 3*emb_sz) embedding for one document (inference.py:74-93)."""
+        ids = self.numericalize(text)
+        if not ids:
+            ids = [self.vocab.stoi.get("xxunk", 0)]
+        t = torch.tensor([ids], dtype=torch.int64, device=self.device)
+        lens = torch.tensor([len(ids)], device=self.device)
+        return self._encode_batch(t, lens).cpu()
+
+    def df_to_embedding(self, dataframe, bs: int = 100) -> np.ndarray:
+        """Bulk path (inference.py:138-229): build docs from (title, body)
+        rows, sort by length, pad per batch, encode, unsort. OOM -> bs//2."""
+        texts = [self.process_dict({"title": t, "body": b})["text"]
+                 for t, b in zip(dataframe["title"], dataframe["body"])]
+        return self.texts_to_embedding(texts, bs=bs)
+
+    def texts_to_embedding(self, texts: Sequence[str], bs: int = 100) -> np.ndarray:
+        docs = [self.numericalize(t) or [0] for t in texts]
+        order = sorted(range(len(docs)), key=lambda i: len(docs[i]))
+        out = np.empty((len(docs), 3 * self.emb_sz), dtype=np.float32)
+        i = 0
+        while i < len(order):
+            cur_bs = bs
+            while True:
+                idxs = order[i: i + cur_bs]
+                batch_docs = [docs[j] for j in idxs]
+                T = max(len(d) for d in batch_docs)
+                ids = torch.full((len(idxs), T), self.pad_idx, dtype=torch.int64)
+                for r, d in enumerate(batch_docs):
+                    ids[r, :len(d)] = torch.tensor(d)
+                lens = torch.tensor([len(d) for d in batch_docs])
+                try:
+                    pooled = self._encode_batch(ids.to(self.device),
+                                                lens.to(self.device))
+                    break
+                except torch.cuda.OutOfMemoryError:
+                    # reference behavior: halve the batch until it fits
+                    # (inference.py:214-223)
+                    torch.cuda.empty_cache()
+                    cur_bs //= 2
+                    if cur_bs < 1:
+                        raise
+            for r, j in enumerate(idxs):
+                out[j] = pooled[r].cpu().numpy()
+            i += len(idxs)
+        return out
+
+    # alias matching the flask-app copy (Issue_Embeddings/flask_app/inference.py)
+    df_to_emb = df_to_embedding
+
+    def batch_seq_pool(self, hidden: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
+        return concat_pool(hidden, lengths)
+
+
+def save_artifacts(model: AWDLSTM, vocab: Vocab, path) -> None:
+    """Write the artifact layout InferenceWrapper loads."""
+    root = Path(path)
+    root.mkdir(parents=True, exist_ok=True)
+    enc = model.encoder
+    (root / "config.json").write_text(json.dumps({
+        "emb_sz": enc.emb_sz, "n_hid": enc.n_hid, "n_layers": enc.n_layers,
+        "encoder_file": "encoder.pth"}))
+    vocab.save(root / "vocab.json")
+    model.save_encoder(root / "encoder.pth")
