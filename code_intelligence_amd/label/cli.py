@@ -1,0 +1,71 @@
+"""Dev CLI (reference: py/label_microservice/cli.py): fetch an issue as
+JSON, publish a label request to the queue, pretty-print worker JSONL logs.
+
+  python -m code_intelligence_amd.label.cli get-issue --url ...
+  python -m code_intelligence_amd.label.cli label-issue --issue owner/repo#123
+  python -m code_intelligence_amd.label.cli logs --path worker.jsonl
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+from ..gh.graphql import GraphQLClient
+from ..gh import github_util
+from ..gh.util import parse_issue_spec, build_issue_url
+from .queueing import LocalQueue
+
+
+def cmd_get_issue(args):
+    client = GraphQLClient(token=args.token)
+    issue = github_util.get_issue(args.url, client)
+    print(json.dumps(issue, indent=2))
+
+
+def cmd_label_issue(args):
+    spec = parse_issue_spec(args.issue)
+    if spec is None:
+        sys.exit(f"bad issue spec: {args.issue}")
+    owner, repo, num = spec
+    q = LocalQueue(spool_path=args.spool)
+    mid = q.publish(repo_owner=owner, repo_name=repo, issue_num=num,
+                    installation_id=args.installation_id or "")
+    print(f"published {mid} -> {build_issue_url(owner, repo, num)}")
+
+
+def cmd_logs(args):
+    with open(args.path) as f:
+        for line in f:
+            try:
+                obj = json.loads(line)
+            except json.JSONDecodeError:
+                print(line, end="")
+                continue
+            t = obj.get("time", "")
+            lvl = obj.get("level", "")
+            print(f"{t} {lvl:8} {obj.get('message', '')} "
+                  f"{ {k: v for k, v in obj.items() if k not in ('time', 'level', 'message', 'filename', 'line_number', 'thread')} }")
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    sub = p.add_subparsers(dest="cmd", required=True)
+    g = sub.add_parser("get-issue")
+    g.add_argument("--url", required=True)
+    g.add_argument("--token", default=None)
+    g.set_defaults(fn=cmd_get_issue)
+    l = sub.add_parser("label-issue")
+    l.add_argument("--issue", required=True, help="owner/repo#num")
+    l.add_argument("--spool", default="/tmp/ci_queue_spool.jsonl")
+    l.add_argument("--installation_id", default=None)
+    l.set_defaults(fn=cmd_label_issue)
+    lg = sub.add_parser("logs")
+    lg.add_argument("--path", required=True)
+    lg.set_defaults(fn=cmd_logs)
+    args = p.parse_args(argv)
+    args.fn(args)
+
+
+if __name__ == "__main__":
+    main()

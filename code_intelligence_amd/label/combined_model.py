@@ -1,0 +1,29 @@
+"""Ensemble: run member models serially, keep per-label max probability
+(reference: py/label_microservice/combined_model.py:15-54)."""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+from .models import IssueLabelModel
+
+
+class CombinedLabelModels(IssueLabelModel):
+    def __init__(self, models: List[IssueLabelModel]):
+        self.models = models
+
+    def predict_issue_labels(self, org: str, repo: str, title: str,
+                             text: List[str], context: Optional[dict] = None
+                             ) -> Dict[str, float]:
+        preds = []
+        for m in self.models:
+            preds.append(m.predict_issue_labels(org, repo, title, text, context))
+        return self._combine_predictions(preds)
+
+    @staticmethod
+    def _combine_predictions(predictions: List[Dict[str, float]]) -> Dict[str, float]:
+        out: Dict[str, float] = {}
+        for p in predictions:
+            for label, prob in (p or {}).items():
+                if label not in out or prob > out[label]:
+                    out[label] = prob
+        return out

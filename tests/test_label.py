@@ -1,0 +1,272 @@
+"""Label-microservice tests (offline fakes — reference test techniques,
+SURVEY.md §4: mocked prediction clients, fake embedding service, crafted
+threshold datasets)."""
+import numpy as np
+import pytest
+import torch
+import yaml
+
+from code_intelligence_amd.label.automl_model import AutoMLModel
+from code_intelligence_amd.label.combined_model import CombinedLabelModels
+from code_intelligence_amd.label.issue_label_predictor import IssueLabelPredictor
+from code_intelligence_amd.label.mlp import MLPHead, MLPWrapper
+from code_intelligence_amd.label.models import IssueLabelModel
+from code_intelligence_amd.label.queueing import LocalQueue
+from code_intelligence_amd.label.repo_config import RepoConfig
+from code_intelligence_amd.label.repo_specific_model import RepoSpecificLabelModel
+from code_intelligence_amd.label.universal_kind_label_model import (
+    UniversalKindLabelModel)
+from code_intelligence_amd.label.worker import Worker
+
+
+class StaticModel(IssueLabelModel):
+    def __init__(self, preds):
+        self.preds = preds
+
+    def predict_issue_labels(self, org, repo, title, text, context=None):
+        return dict(self.preds)
+
+
+# --- MLP wrapper --------------------------------------------------------
+def test_mlp_fit_and_predict_separable():
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(400, 8)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    w = MLPWrapper(in_dim=8, hidden=(16,), n_labels=1, lr=1e-2)
+    w.fit(X, y)
+    auc = w.calculate_auc(X, y)
+    assert auc > 0.95, auc
+
+
+def test_threshold_search_excludes_unsatisfiable_labels():
+    """reference test_mlp.py:8-59: labels whose P-R curve can't reach
+    precision>=0.7 & recall>=0.5 get threshold None (never predict)."""
+    w = MLPWrapper(in_dim=4, hidden=(8,), n_labels=3)
+
+    class FixedProb(MLPWrapper):
+        pass
+
+    X = np.zeros((6, 4), dtype=np.float32)
+    y = np.array([[1, 0, 1], [1, 0, 0], [1, 0, 1],
+                  [0, 1, 0], [0, 1, 1], [0, 1, 0]], dtype=np.float32)
+    probs = np.array([[.9, .1, .5], [.8, .2, .5], [.85, .15, .5],
+                      [.1, .2, .5], [.2, .1, .5], [.15, .3, .5]],
+                     dtype=np.float32)
+    w.predict_probabilities = lambda X_: probs  # type: ignore
+    thr = w.find_probability_thresholds(X, y)
+    assert thr[0] is not None        # label 0 perfectly separable
+    assert thr[1] is None            # label 1: probs anti-correlated
+    # label 2: constant 0.5 prob, precision 0.5 < 0.7 -> None
+    assert thr[2] is None
+
+
+def test_mlp_save_load_roundtrip(tmp_path):
+    w = MLPWrapper(in_dim=4, hidden=(8,), n_labels=2)
+    w.probability_thresholds = {0: 0.6, 1: None}
+    p = tmp_path / "m.dpkl"
+    w.save_model(p)
+    w2 = MLPWrapper.load_model(p)
+    assert w2.probability_thresholds == {0: 0.6, 1: None}
+    X = np.random.default_rng(0).normal(size=(3, 4)).astype(np.float32)
+    assert np.allclose(w.predict_probabilities(X), w2.predict_probabilities(X))
+
+
+# --- combined model (reference combined_model_test.py) ------------------
+def test_combined_max_merge():
+    m = CombinedLabelModels([StaticModel({"bug": 0.9, "area/ops": 0.4}),
+                             StaticModel({"bug": 0.7, "feature": 0.8})])
+    preds = m.predict_issue_labels("o", "r", "t", [])
+    assert preds == {"bug": 0.9, "area/ops": 0.4, "feature": 0.8}
+
+
+# --- automl model (reference automl_model_test.py) ----------------------
+def test_automl_threshold_and_dash_mapping():
+    m = AutoMLModel("m", predict_fn=lambda doc: [
+        ("kind-bug", 0.9), ("area-platform-gcp", 0.6), ("lowconf", 0.2)])
+    preds = m.predict_issue_labels("org", "repo", "title", ["text"])
+    # 0.5 threshold filters lowconf; '-' -> '/' mapped ONCE
+    assert preds == {"kind/bug": 0.9, "area/platform-gcp": 0.6}
+
+
+# --- repo-specific model (reference repo_specific_model_test.py) --------
+class FakeEmbeddingSession:
+    def __init__(self, status=200, dim=2400):
+        self.status = status
+        self.dim = dim
+
+    def post(self, url, json=None, **kw):
+        class R:
+            pass
+        r = R()
+        r.status_code = self.status
+        vec = np.arange(self.dim, dtype="<f4")
+        r.content = vec.tobytes()
+        return r
+
+
+def test_repo_specific_predicts_with_thresholds():
+    w = MLPWrapper(in_dim=1600, hidden=(4,), n_labels=2)
+    w.predict_probabilities = lambda X: np.array([[0.2, 0.9]])  # type: ignore
+    m = RepoSpecificLabelModel(w, ["bug", "feature"], {0: 0.5, 1: 0.5},
+                               session=FakeEmbeddingSession())
+    preds = m.predict_issue_labels("o", "r", "t", ["b"])
+    assert preds == {"feature": pytest.approx(0.9)}
+
+
+def test_repo_specific_embedding_service_down_returns_empty():
+    w = MLPWrapper(in_dim=1600, hidden=(4,), n_labels=1)
+    m = RepoSpecificLabelModel(w, ["bug"], {0: 0.5},
+                               session=FakeEmbeddingSession(status=404))
+    assert m.predict_issue_labels("o", "r", "t", ["b"]) == {}
+
+
+def test_repo_specific_truncates_to_1600():
+    captured = {}
+    w = MLPWrapper(in_dim=1600, hidden=(4,), n_labels=1)
+
+    def capture(X):
+        captured["shape"] = X.shape
+        return np.array([[0.9]])
+    w.predict_probabilities = capture  # type: ignore
+    m = RepoSpecificLabelModel(w, ["bug"], {0: 0.5},
+                               session=FakeEmbeddingSession(dim=2400))
+    m.predict_issue_labels("o", "r", "t", [])
+    assert captured["shape"] == (1, 1600)  # embeddings.py:116 truncation
+
+
+def test_repo_config_paths():
+    c = RepoConfig("kubeflow", "examples")
+    assert c.model_gcs_uri == "gs://repo-models/kubeflow/examples.model.dpkl"
+    assert c.labels_gcs_uri == "gs://repo-models/kubeflow/examples.labels.yaml"
+
+
+def test_repo_specific_from_repo_loads_artifacts(tmp_path):
+    from code_intelligence_amd.gh.gcs_util import ObjectStore
+    store = ObjectStore(root=tmp_path)
+    w = MLPWrapper(in_dim=1600, hidden=(4,), n_labels=2)
+    w.probability_thresholds = {0: 0.5, 1: None}
+    local = tmp_path / "m.dpkl"
+    w.save_model(local)
+    cfg = RepoConfig("o", "r")
+    store.upload(str(local), cfg.model_gcs_uri)
+    store.write_bytes(cfg.labels_gcs_uri, yaml.safe_dump(
+        {"labels": ["bug", "feature"],
+         "probability_thresholds": {0: 0.5, 1: None}}).encode())
+    m = RepoSpecificLabelModel.from_repo("o", "r", store=store,
+                                         session=FakeEmbeddingSession())
+    assert m.label_names == ["bug", "feature"]
+    assert m.thresholds == {0: 0.5, 1: None}
+
+
+# --- universal model ----------------------------------------------------
+def test_universal_model_thresholds(tmp_path):
+    m = UniversalKindLabelModel()
+    # force logits so sigmoid probs are deterministic
+    with torch.no_grad():
+        m.net.out.weight.zero_()
+        m.net.out.bias.copy_(torch.tensor([2.0, -2.0, 0.35]))
+    preds = m.predict_issue_labels("o", "r", "crash", ["boom"])
+    assert "bug" in preds            # sigmoid(2.0)=0.88 >= 0.52
+    assert "feature" not in preds    # 0.12 < 0.52
+    assert "question" not in preds   # sigmoid(0.35)=0.587 < 0.60
+    m.save(tmp_path / "u")
+    m2 = UniversalKindLabelModel.load(tmp_path / "u")
+    assert m2.predict_issue_labels("o", "r", "crash", ["boom"]).keys() == preds.keys()
+
+
+# --- predictor routing (issue_label_predictor.py:146-155) ---------------
+def test_predictor_routing_specificity():
+    p = IssueLabelPredictor(model_config={}, universal=StaticModel({"u": 0.9}))
+    p.models["kubeflow_combined"] = StaticModel({"org": 0.9})
+    p.models["kubeflow/examples_combined"] = StaticModel({"repo": 0.9})
+    assert p.predict_labels_for_data("kubeflow", "examples", "t", []) == {"repo": 0.9}
+    assert p.predict_labels_for_data("kubeflow", "other", "t", []) == {"org": 0.9}
+    assert p.predict_labels_for_data("x", "y", "t", []) == {"u": 0.9}
+
+
+def test_predictor_payload_dispatch():
+    p = IssueLabelPredictor(model_config={}, universal=StaticModel({"bug": 0.8}))
+    preds = p.predict({"repo_owner": "o", "repo_name": "r",
+                       "title": "t", "text": ["b"]})
+    assert preds == {"bug": 0.8}
+    with pytest.raises(ValueError):
+        p.predict({"title": "no repo"})
+
+
+# --- worker end-to-end (fakes) ------------------------------------------
+class RecordingGitHub:
+    def __init__(self, existing_comments=None):
+        self.labels = []
+        self.comments = []
+        self.existing = existing_comments or []
+
+    def add_labels(self, owner, repo, num, labels):
+        self.labels.append((owner, repo, num, labels))
+
+    def add_comment(self, owner, repo, num, body):
+        self.comments.append(body)
+
+    def list_comments(self, owner, repo, num):
+        return self.existing
+
+
+def _mk_worker(predictions, repo_cfg=None, issue_data=None, gh=None):
+    pred = IssueLabelPredictor(model_config={},
+                               universal=StaticModel(predictions))
+    gh = gh or RecordingGitHub()
+    w = Worker(queue=LocalQueue(), predictor=pred, github=gh,
+               repo_config_fn=lambda o, r: repo_cfg)
+    return w, gh
+
+
+def test_worker_applies_labels_and_comments():
+    w, gh = _mk_worker({"bug": 0.9})
+    added = w.add_labels_to_issue("o", "r", 1, {"bug": 0.9},
+                                  issue_data={"labels": [], "removed_labels": []})
+    assert added == ["bug"]
+    assert gh.labels == [("o", "r", 1, ["bug"])]
+    assert "| bug | 0.90 |" in gh.comments[0]
+
+
+def test_worker_dedupes_existing_and_removed():
+    w, gh = _mk_worker({})
+    added = w.add_labels_to_issue(
+        "o", "r", 1, {"bug": 0.9, "feature": 0.8},
+        issue_data={"labels": ["bug"], "removed_labels": ["feature"]})
+    assert added == []           # human removed 'feature'; 'bug' already there
+    assert gh.labels == []
+
+
+def test_worker_alias_and_allowlist():
+    cfg = {"label-alias": {"bug": "kind/bug"},
+           "predicted-labels": ["kind/bug"]}
+    out = Worker.apply_repo_config(cfg, {"bug": 0.9, "feature": 0.8})
+    assert out == {"kind/bug": 0.9}
+
+
+def test_worker_callback_acks_even_on_failure():
+    from code_intelligence_amd.label.queueing import Message
+    w, gh = _mk_worker({"bug": 0.9})
+    msg = Message(attributes={})  # missing repo_owner -> KeyError inside
+    w.callback(msg)
+    assert msg._acked  # always ack (worker.py:231)
+
+
+def test_worker_skips_duplicate_comment():
+    gh = RecordingGitHub(existing_comments=[
+        {"body": Worker.BOT_MARKER + " earlier"}])
+    w, _ = _mk_worker({"bug": 0.9}, gh=gh)
+    w.add_labels_to_issue("o", "r", 1, {"bug": 0.9},
+                          issue_data={"labels": [], "removed_labels": []})
+    assert gh.labels  # label applied
+    assert gh.comments == []  # no second comment
+
+
+def test_local_queue_spool_roundtrip(tmp_path):
+    spool = tmp_path / "spool.jsonl"
+    prod = LocalQueue(spool_path=str(spool))
+    prod.publish(repo_owner="o", repo_name="r", issue_num=3)
+    cons = LocalQueue(spool_path=str(spool))
+    msg = cons.pull(timeout=0.2)
+    assert msg is not None
+    assert msg.attributes["issue_num"] == "3"
