@@ -17,9 +17,17 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
   const T* x = logits + (long)row * row_stride;
   __shared__ float red[NW];
   __shared__ float bcast;
-  // pass 1: global row max
+  // pass 1: global row max (16-B vector loads, G13)
+  constexpr int VEC = 16 / sizeof(T);
+  const int Vv = V / VEC * VEC;
   float mx = -3.4e38f;
-  for (int v = threadIdx.x; v < V; v += THREADS) mx = fmaxf(mx, ld(x + v));
+  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    T buf[VEC];
+    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) mx = fmaxf(mx, ld(buf + e));
+  }
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) mx = fmaxf(mx, ld(x + v));
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1)
     mx = fmaxf(mx, __shfl_down(mx, off));
@@ -35,7 +43,13 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
   __syncthreads();  // red[] reused below
   // pass 2: sum exp(x - max)
   float s = 0.f;
-  for (int v = threadIdx.x; v < V; v += THREADS) s += __expf(ld(x + v) - mx);
+  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    T buf[VEC];
+    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) s += __expf(ld(buf + e) - mx);
+  }
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) s += __expf(ld(x + v) - mx);
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
   if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = s;
@@ -62,32 +76,49 @@ void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
 }
 
 // in-place: logits <- (exp(logits - lse) - onehot) * scale   (scale on device)
-template <typename T>
-__global__ void ce_dlogits_kernel(T* __restrict__ logits,
+// one block per row, vectorized 8-wide in-place transform (G13).
+template <typename T, int THREADS>
+__global__ void ce_dlogits_kernel(T* __restrict__ logits, long row_stride,
                                   const long* __restrict__ targets,
                                   const float* __restrict__ lse,
                                   const float* __restrict__ scale,
-                                  int N, int V) {
-  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= (long)N * V) return;
-  const int row = idx / V, col = idx % V;
-  float p = __expf(ld(logits + idx) - lse[row]);
-  if (col == (int)targets[row]) p -= 1.f;
-  st(logits + idx, p * scale[0]);
+                                  int V) {
+  const int row = blockIdx.x;
+  T* x = logits + (long)row * row_stride;
+  const float l = lse[row];
+  const float sc = scale[0];
+  const int tgt = (int)targets[row];
+  constexpr int VEC = 16 / sizeof(T);  // one 16-B vector per thread-step
+  const int Vv = V / VEC * VEC;
+  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    T buf[VEC];
+    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      float p = __expf(ld(buf + e) - l);
+      if (v + e == tgt) p -= 1.f;
+      st(buf + e, p * sc);
+    }
+    *reinterpret_cast<int4*>(x + v) = *reinterpret_cast<const int4*>(buf);
+  }
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
+    float p = __expf(ld(x + v) - l);
+    if (v == tgt) p -= 1.f;
+    st(x + v, p * sc);
+  }
 }
 
 void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor lse,
                 at::Tensor scale) {
   CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
   const int N = logits.size(0), V = logits.size(1);
-  const int threads = 256;
-  const long total = (long)N * V;
+  constexpr int THREADS = 256;
   CI_DISPATCH_FB(logits.scalar_type(), "ce_dlogits", [&] {
-    hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t>),
-        dim3(ceil_div(total, threads)), dim3(threads), 0, stream(),
-        reinterpret_cast<scalar_t*>(logits.data_ptr()),
+    hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t, THREADS>),
+        dim3(N), dim3(THREADS), 0, stream(),
+        reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
         targets.data_ptr<long>(), lse.data_ptr<float>(),
-        scale.data_ptr<float>(), N, V);
+        scale.data_ptr<float>(), V);
   });
 }
 

@@ -1,27 +1,31 @@
-// K2 fused mode: hand-written CDNA4 MFMA LSTM cell kernel.
+// K2 fused mode: hand-written CDNA4 MFMA LSTM cell kernel (v2).
 //
-// Per timestep, ONE kernel computes
-//   pre[b, n'] = sum_k h_prev[b,k] * W_hh[row(n'), k]  (+ xp + bias)
-// with the output columns GATE-INTERLEAVED: n' = 4*j + g maps to the
-// original gate row g*H + j (g in {i,f,g,o}, PyTorch gate order). That puts
-// all four gates of hidden unit j in four adjacent columns of the same
-// output tile, so the epilogue can finish the whole cell locally:
-//   c_t = sigm(f)*c_{t-1} + sigm(i)*tanh(g);  h_t = sigm(o)*tanh(c_t)
-// and store h (bf16), c (fp32) and the post-activation gates (for K7
-// backward) without a second kernel or a round-trip of the 4H-wide
-// pre-activation matrix through HBM.
+// Per timestep, ONE kernel computes the recurrent GEMM
+//   pre[b, n'] = sum_k h_prev[b,k] * W_hh[row(n'), k]
+// with output columns GATE-INTERLEAVED (n' = 4*j + g -> original gate row
+// g*H + j, PyTorch gate order i,f,g,o) so all four gates of hidden unit j
+// land in adjacent columns of one tile and the epilogue finishes the cell
+// locally: c_t = sigm(f)*c + sigm(i)*tanh(g); h_t = sigm(o)*tanh(c_t),
+// storing h (bf16), c (fp32) and post-activation gates (for K7 backward)
+// with no extra kernel and no HBM round-trip of the 4H-wide pre-matrix.
 //
-// GEMM structure (cdna_hip_programming.md §5, "step-2/3" class):
-//   128x128 tile, BK=64, 4 waves of 64x64, mfma_f32_16x16x32_bf16,
-//   double-buffered LDS staging with XOR-swizzled 16B chunks
-//   (T2: byte ^= (row&7)<<4 equivalent) read back as b128 fragments.
-// The interleave permutation row(n') = (n'&3)*H + (n'>>2) is applied on the
-// *global source address* of the W staging loads — W_hh itself stays in the
-// checkpoint layout, no pre-permute pass.
+// v2 structure (cdna_hip_programming.md par.5 "step-3" class, adapted):
+//  * 128x64 tile, BK=64, 4 waves of 64x32, mfma_f32_16x16x32_bf16
+//    -> 600 blocks at the deployed shape (B=512, H=2400): ~2.3 blocks/CU
+//    (the v1 128x128 tile gave only 300 blocks = 1.17/CU, latency-bound)
+//  * interior tiles staged by __builtin_amdgcn_global_load_lds (16 B,
+//    lane-linear LDS dest; the T2 XOR swizzle moves to the per-lane SOURCE
+//    address), double-buffered, issued BEFORE the MFMAs of the current
+//    tile so the DMA flight hides under compute
+//  * edge tiles (batch/N/K tails) fall back to register staging with the
+//    same swizzled LDS image
+//  * XCD-aware bijective blockIdx remap (T1): the MT blocks sharing one
+//    W-panel run on one XCD so the panel stays in that XCD's L2
+//  * the W interleave permutation row(n') = (n'&3)*H + (n'>>2) is applied
+//    on the staging source address; W_hh stays in checkpoint layout.
 //
-// Reference op semantics: SURVEY.md §2.4 K2/K3; the weight-drop mask is
-// applied by the caller (masked W_hh is what arrives here), matching
-// fastai WeightDropout (train.py:70).
+// Weight-drop (K3) masks are applied by the caller (masked W arrives
+// here), matching fastai WeightDropout (train.py:70).
 #include "common.h"
 
 namespace ci {
@@ -29,51 +33,72 @@ namespace ci {
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-constexpr int BM = 128, BN = 128, BK = 64;
-constexpr int THREADS = 256;            // 4 waves, each owns a 64x64 subtile
-constexpr int LDS_ELEMS = BM * BK;      // per operand tile (bf16 elements)
+constexpr int BM = 128, BN = 64, BK = 64;
+constexpr int THREADS = 256;             // 4 waves: 2(M) x 2(N) of 64x32
+constexpr int A_ELEMS = BM * BK;         // bf16 elements per A tile
+constexpr int B_ELEMS = BN * BK;
 
-// element offset of (row, col) in a swizzled [rows][BK] bf16 LDS tile:
-// 16B chunks within a 128B row are XOR'd by (row&7) — T2 swizzle.
+// element offset of (row, col) in the swizzled [rows][BK] bf16 LDS tile:
+// 16B chunks within a 128B row are XOR'd by (row&7) (T2).
 static __device__ __forceinline__ int swz(int row, int col) {
   const int chunk = col >> 3;
   return row * BK + ((chunk ^ (row & 7)) << 3) + (col & 7);
 }
 
-// stage a BMxBK bf16 tile from global (row-major, arbitrary row stride,
-// optional row permutation for W) into swizzled LDS. 256 threads, each
-// moves 4 x 16B chunks. rows beyond row_lim / k beyond k_lim are zeroed.
 template <bool PERM>
-static __device__ __forceinline__ void stage_tile(
+static __device__ __forceinline__ long src_row(int abs_row, int H) {
+  return PERM ? (long)(abs_row & 3) * H + (abs_row >> 2) : abs_row;
+}
+
+// ---- glds staging: linear LDS dest, swizzle folded into the source ----
+// R rows x 64 cols bf16 = R*8 16B slots; one wave-instruction covers 64
+// slots (lane-linear). slot s16 -> row = s16>>3, chunk = (s16&7) ^ (row&7).
+template <int R, bool PERM>
+static __device__ __forceinline__ void stage_glds(
     const __hip_bfloat16* __restrict__ src, long row_stride, int row0,
-    int row_lim, int k0, int k_lim, int H, __hip_bfloat16* lds) {
-  const int tid = threadIdx.x;
-  // thread t: row pair r = t>>1, half = t&1 covers 4 chunks of 8 elems
-  const int r = tid >> 1;
-  const int cbase = (tid & 1) * 4;  // chunk index base (of 8 per row)
-  const int grow_t = row0 + r;
-  long grow;
-  if (PERM) {
-    // output col n' = grow_t ; source row = (n'&3)*H + (n'>>2)
-    grow = (long)(grow_t & 3) * H + (grow_t >> 2);
-  } else {
-    grow = grow_t;
-  }
-  const bool rok = grow_t < row_lim;
+    int k0, int H, __hip_bfloat16* lds) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
   #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    const int chunk = cbase + c;
-    const int k = k0 + (chunk << 3);
-    bf16x8 v = {};
-    if (rok && k + 8 <= k_lim) {
-      v = *reinterpret_cast<const bf16x8*>(src + grow * row_stride + k);
-    }
-    *reinterpret_cast<bf16x8*>(lds + swz(r, chunk << 3)) = v;
+  for (int i = 0; i < R / 32; ++i) {
+    const int s16 = i * 256 + wave * 64 + lane;
+    const int row = s16 >> 3;
+    const int chunk = (s16 & 7) ^ (row & 7);
+    const long grow = src_row<PERM>(row0 + row, H);
+    const __hip_bfloat16* g = src + grow * row_stride + k0 + chunk * 8;
+    // C-style casts switch address space (generic->global/LDS) — the
+    // pattern ck_tile uses for llvm.amdgcn.*.load.lds operands
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)(g),
+        (__attribute__((address_space(3))) unsigned int*)(
+            lds + (long)(i * 256 + wave * 64) * 8),
+        16, 0, 0);
   }
 }
 
-// fragment loads: lane l reads rows (fr*16 + (l&15)), k ((l>>4)*8) of the
-// 64-wide k-slice ks (0 or 1) — 16B contiguous = ds_read_b128.
+// ---- register staging fallback for edge tiles (zero-filled) -----------
+template <int R, bool PERM>
+static __device__ __forceinline__ void stage_reg(
+    const __hip_bfloat16* __restrict__ src, long row_stride, int row0,
+    int row_lim, int k0, int k_lim, int H, __hip_bfloat16* lds) {
+  // 256 threads move R*8 slots; thread t handles slots t, t+256, ...
+  #pragma unroll
+  for (int i = 0; i < R / 32; ++i) {
+    const int s16 = i * 256 + threadIdx.x;
+    const int row = s16 >> 3;
+    const int x = s16 & 7;
+    const int chunk = x ^ (row & 7);
+    const int abs_row = row0 + row;
+    const int k = k0 + chunk * 8;
+    bf16x8 v = {};
+    if (abs_row < row_lim && k + 8 <= k_lim) {
+      v = *reinterpret_cast<const bf16x8*>(
+          src + src_row<PERM>(abs_row, H) * row_stride + k);
+    }
+    *reinterpret_cast<bf16x8*>(lds + (long)s16 * 8) = v;
+  }
+}
+
 static __device__ __forceinline__ bf16x8 frag(const __hip_bfloat16* lds,
                                               int row_base, int ks, int lane) {
   const int row = row_base + (lane & 15);
@@ -91,87 +116,98 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
     float* __restrict__ c_out, long co_rs,
     __hip_bfloat16* __restrict__ gates_out, long go_rs,  // block layout
     int B, int H, int MT) {
-  const int NT = gridDim.x / MT;
-  // block -> (mt, nt): consecutive ids share nt so the W panel stays hot in
-  // one XCD's L2 (placement is a perf hint only).
-  const int nt = blockIdx.x / MT;
-  const int mt = blockIdx.x % MT;
+  // T1 bijective XCD remap: blocks sharing a W panel -> one XCD's L2
+  const int nwg = gridDim.x;
+  const int p = blockIdx.x;
+  const int xcd = p % 8, pos = p / 8;
+  const int q = nwg / 8, r = nwg % 8;
+  const int logical = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  const int nt = logical / MT;
+  const int mt = logical % MT;
+
   const int N = 4 * H, K = H;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave & 1) * 64;    // wave row offset in tile
-  const int wn = (wave >> 1) * 64;   // wave col offset
+  const int wm = (wave & 1) * 64;    // wave row offset (2 waves over 128)
+  const int wn = (wave >> 1) * 32;   // wave col offset (2 waves over 64)
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __hip_bfloat16* lsb = reinterpret_cast<__hip_bfloat16*>(smem);
-  // buf 0: [A0 | B0], buf 1: [A1 | B1]
-#define LA(buf) (lsb + (buf) * 2 * LDS_ELEMS)
-#define LB(buf) (lsb + (buf) * 2 * LDS_ELEMS + LDS_ELEMS)
+  // buffer b: [A_b (128x64) | B_b (64x64)]
+#define LA(b) (lsb + (b) * (A_ELEMS + B_ELEMS))
+#define LB(b) (lsb + (b) * (A_ELEMS + B_ELEMS) + A_ELEMS)
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[4][2] = {};
 
   const int row0_a = mt * BM, row0_b = nt * BN;
   const int nk = (K + BK - 1) / BK;
-  stage_tile<false>(h_prev, h_rs, row0_a, B, 0, K, H, LA(0));
-  stage_tile<true>(w_hh, K, row0_b, N, 0, K, H, LB(0));
+  const bool interior_rows = (row0_a + BM <= B) && (row0_b + BN <= N);
+
+  auto stage = [&](int kt, int buf) {
+    const int k0 = kt * BK;
+    if (interior_rows && k0 + BK <= K) {
+      stage_glds<BM, false>(h_prev, h_rs, row0_a, k0, H, LA(buf));
+      stage_glds<BN, true>(w_hh, K, row0_b, k0, H, LB(buf));
+    } else {
+      stage_reg<BM, false>(h_prev, h_rs, row0_a, B, k0, K, H, LA(buf));
+      stage_reg<BN, true>(w_hh, K, row0_b, N, k0, K, H, LB(buf));
+    }
+  };
+
+  stage(0, 0);
   __syncthreads();
 
   for (int kt = 0; kt < nk; ++kt) {
     const int cur = kt & 1;
-    if (kt + 1 < nk) {
-      stage_tile<false>(h_prev, h_rs, row0_a, B, (kt + 1) * BK, K, H, LA(cur ^ 1));
-      stage_tile<true>(w_hh, K, row0_b, N, (kt + 1) * BK, K, H, LB(cur ^ 1));
-    }
+    if (kt + 1 < nk) stage(kt + 1, cur ^ 1);  // async: flight hides under MFMA
     #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      bf16x8 af[4], bf[4];
+      bf16x8 af[4], bfr[2];
       #pragma unroll
       for (int f = 0; f < 4; ++f) af[f] = frag(LA(cur), wm + f * 16, ks, lane);
       #pragma unroll
-      for (int f = 0; f < 4; ++f) bf[f] = frag(LB(cur), wn + f * 16, ks, lane);
+      for (int f = 0; f < 2; ++f) bfr[f] = frag(LB(cur), wn + f * 16, ks, lane);
       #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
         #pragma unroll
-        for (int fn = 0; fn < 4; ++fn)
+        for (int fn = 0; fn < 2; ++fn)
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[fm], bf[fn], acc[fm][fn], 0, 0, 0);
+              af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
     }
-    __syncthreads();
+    __syncthreads();  // drains the glds queue (vmcnt0) + barrier
   }
 
-  // ---- epilogue: stash pre-activations in LDS, finish the cell ----------
-  // fp32 tile [BM][BN] = 64 KiB (fits the default dynamic-LDS cap; the
-  // 2-way b32 write conflict this leaves is epilogue-only and cheap)
-  float* pre = reinterpret_cast<float*>(smem);
+  // ---- epilogue: stash pre-activations in LDS, finish the cell ---------
+  float* pre = reinterpret_cast<float*>(smem);   // [BM][BN] fp32 = 32 KiB
   constexpr int PRS = BN;
   #pragma unroll
   for (int fm = 0; fm < 4; ++fm) {
     #pragma unroll
-    for (int fn = 0; fn < 4; ++fn) {
+    for (int fn = 0; fn < 2; ++fn) {
       const int col = wn + fn * 16 + (lane & 15);
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = wm + fm * 16 + ((lane >> 4) << 2) + r;
-        pre[row * PRS + col] = acc[fm][fn][r];
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = wm + fm * 16 + ((lane >> 4) << 2) + rr;
+        pre[row * PRS + col] = acc[fm][fn][rr];
       }
     }
   }
   __syncthreads();
 
-  // 128 rows x 32 units per tile; thread p handles (b, j) pairs
-  const int jt = nt * 32;  // first hidden unit of this tile
-  for (int p = threadIdx.x; p < BM * 32; p += THREADS) {
-    const int br = p >> 5;           // row within tile
-    const int jj = p & 31;           // unit within tile
+  // BN=64 cols = 16 hidden units; thread p2 handles (b, j) pairs
+  const int jt = nt * (BN / 4);
+  for (int p2 = threadIdx.x; p2 < BM * (BN / 4); p2 += THREADS) {
+    const int br = p2 >> 4;           // row within tile (BN/4 = 16)
+    const int jj = p2 & 15;           // unit within tile
     const int b = row0_a + br;
     const int j = jt + jj;
     if (b >= B || j >= H) continue;
-    const float* q = pre + br * PRS + jj * 4;
+    const float* qp = pre + br * PRS + jj * 4;
     const long xo = (long)b * xp_rs + j;
-    float gi = q[0] + __bfloat162float(xp[xo]) + bias[j];
-    float gf = q[1] + __bfloat162float(xp[xo + H]) + bias[j + H];
-    float gg = q[2] + __bfloat162float(xp[xo + 2 * H]) + bias[j + 2 * H];
-    float go = q[3] + __bfloat162float(xp[xo + 3 * H]) + bias[j + 3 * H];
+    float gi = qp[0] + __bfloat162float(xp[xo]) + bias[j];
+    float gf = qp[1] + __bfloat162float(xp[xo + H]) + bias[j + H];
+    float gg = qp[2] + __bfloat162float(xp[xo + 2 * H]) + bias[j + 2 * H];
+    float go = qp[3] + __bfloat162float(xp[xo + 3 * H]) + bias[j + 3 * H];
     gi = sigmoidf_(gi); gf = sigmoidf_(gf); gg = tanhf(gg); go = sigmoidf_(go);
     const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
     const float h = go * tanhf(c);
@@ -199,8 +235,9 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
   const int MT = ceil_div(B, BM);
   const dim3 grid(MT * ceil_div(4 * H, BN));
-  const size_t lds = std::max((size_t)4 * LDS_ELEMS * sizeof(__hip_bfloat16),
-                              (size_t)BM * BN * sizeof(float));
+  const size_t lds = std::max(
+      (size_t)2 * (A_ELEMS + B_ELEMS) * sizeof(__hip_bfloat16),
+      (size_t)BM * BN * sizeof(float));
   auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
   auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
   auto* gp = reinterpret_cast<__hip_bfloat16*>(gates.data_ptr());
