@@ -14,10 +14,10 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
                        at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
                        at::Tensor dh0, at::Tensor dc0);
 at::Tensor concat_pool(at::Tensor hidden, at::Tensor lengths);
-void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                 at::Tensor tgt);
-void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                at::Tensor scale);
+void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                 at::Tensor lse, at::Tensor tgt);
+void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                at::Tensor lse, at::Tensor scale);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                  std::vector<at::Tensor> masters, std::vector<at::Tensor> eas_,
                  std::vector<at::Tensor> eass, double lr, double b1, double b2,

@@ -10,6 +10,7 @@ namespace ci {
 template <typename T, int THREADS>
 __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride,
                                    const long* __restrict__ targets,
+                                   const float* __restrict__ bias,  // or null
                                    float* __restrict__ lse,
                                    float* __restrict__ tgt, int V) {
   constexpr int NW = THREADS / kWave;
@@ -25,9 +26,11 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
     T buf[VEC];
     *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
     #pragma unroll
-    for (int e = 0; e < VEC; ++e) mx = fmaxf(mx, ld(buf + e));
+    for (int e = 0; e < VEC; ++e)
+      mx = fmaxf(mx, ld(buf + e) + (bias ? bias[v + e] : 0.f));
   }
-  for (int v = Vv + threadIdx.x; v < V; v += THREADS) mx = fmaxf(mx, ld(x + v));
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS)
+    mx = fmaxf(mx, ld(x + v) + (bias ? bias[v] : 0.f));
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1)
     mx = fmaxf(mx, __shfl_down(mx, off));
@@ -47,9 +50,11 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
     T buf[VEC];
     *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
     #pragma unroll
-    for (int e = 0; e < VEC; ++e) s += __expf(ld(buf + e) - mx);
+    for (int e = 0; e < VEC; ++e)
+      s += __expf(ld(buf + e) + (bias ? bias[v + e] : 0.f) - mx);
   }
-  for (int v = Vv + threadIdx.x; v < V; v += THREADS) s += __expf(ld(x + v) - mx);
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS)
+    s += __expf(ld(x + v) + (bias ? bias[v] : 0.f) - mx);
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
   if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = s;
@@ -58,20 +63,23 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
     float tot = 0.f;
     for (int w = 0; w < NW; ++w) tot += red[w];
     lse[row] = mx + __logf(tot);
-    tgt[row] = ld(x + targets[row]);
+    tgt[row] = ld(x + targets[row]) +
+               (bias ? bias[targets[row]] : 0.f);
   }
 }
 
-void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                 at::Tensor tgt) {
+void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                 at::Tensor lse, at::Tensor tgt) {
   CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
   const int N = logits.size(0), V = logits.size(1);
   constexpr int THREADS = 256;
+  const float* bp = bias.numel() ? bias.data_ptr<float>() : nullptr;
   CI_DISPATCH_FB(logits.scalar_type(), "ce_rowstats", [&] {
     hipLaunchKernelGGL((ce_rowstats_kernel<scalar_t, THREADS>), dim3(N),
         dim3(THREADS), 0, stream(),
         reinterpret_cast<const scalar_t*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), lse.data_ptr<float>(), tgt.data_ptr<float>(), V);
+        targets.data_ptr<long>(), bp, lse.data_ptr<float>(),
+        tgt.data_ptr<float>(), V);
   });
 }
 
@@ -80,6 +88,7 @@ void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor lse,
 template <typename T, int THREADS>
 __global__ void ce_dlogits_kernel(T* __restrict__ logits, long row_stride,
                                   const long* __restrict__ targets,
+                                  const float* __restrict__ bias,  // or null
                                   const float* __restrict__ lse,
                                   const float* __restrict__ scale,
                                   int V) {
@@ -95,29 +104,30 @@ __global__ void ce_dlogits_kernel(T* __restrict__ logits, long row_stride,
     *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
     #pragma unroll
     for (int e = 0; e < VEC; ++e) {
-      float p = __expf(ld(buf + e) - l);
+      float p = __expf(ld(buf + e) + (bias ? bias[v + e] : 0.f) - l);
       if (v + e == tgt) p -= 1.f;
       st(buf + e, p * sc);
     }
     *reinterpret_cast<int4*>(x + v) = *reinterpret_cast<const int4*>(buf);
   }
   for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
-    float p = __expf(ld(x + v) - l);
+    float p = __expf(ld(x + v) + (bias ? bias[v] : 0.f) - l);
     if (v == tgt) p -= 1.f;
     st(x + v, p * sc);
   }
 }
 
-void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor lse,
-                at::Tensor scale) {
+void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                at::Tensor lse, at::Tensor scale) {
   CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
   const int N = logits.size(0), V = logits.size(1);
   constexpr int THREADS = 256;
+  const float* bp = bias.numel() ? bias.data_ptr<float>() : nullptr;
   CI_DISPATCH_FB(logits.scalar_type(), "ce_dlogits", [&] {
     hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t, THREADS>),
         dim3(N), dim3(THREADS), 0, stream(),
         reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), lse.data_ptr<float>(),
+        targets.data_ptr<long>(), bp, lse.data_ptr<float>(),
         scale.data_ptr<float>(), V);
   });
 }

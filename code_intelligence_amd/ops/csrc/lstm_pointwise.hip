@@ -128,6 +128,11 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
   const int H = w_hh.size(1);
   auto dc_buf = dcT.clone();                       // (B,H) fp32 running dc
   auto dh_rec = dhT.contiguous();                  // (B,H) scalar running rec grad
+  // one 46 MB transpose buys the NT GEMM layout for all T steps:
+  // mm(dgates, w_hh) NN measured 304 TF vs 503 TF as mm(dgates, w_t.t())
+  // (scripts/gemm_probe.py on MI355X)
+  auto w_hh_tc = w_hh.t().contiguous();
+  auto w_hh_nt = w_hh_tc.t();
   const int threads = 256;
   const int blocks = ceil_div((long)B * H, threads);
   CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
@@ -144,8 +149,8 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
           cs.data_ptr<float>() + (long)t * H, (long)T * H,
           reinterpret_cast<scalar_t*>(dgates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
           B, H);
-      // dh_{t-1} recurrent contribution: dgates_t @ w_hh
-      at::mm_out(dh_rec, dgates.select(1, t), w_hh);
+      // dh_{t-1} recurrent contribution: dgates_t @ w_hh (NT layout)
+      at::mm_out(dh_rec, dgates.select(1, t), w_hh_nt);
     }
   });
   dh0.copy_(dh_rec);
