@@ -230,7 +230,7 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16,
               "fused LSTM cell kernel is bf16; use CI_LSTM_MODE=lib for fp32");
   TORCH_CHECK(w_hh.is_contiguous(), "w_hh must be contiguous");
-  const int B = xp.size(0), T = xp.size(1);
+  const int T = xp.size(0), B = xp.size(1);   // TIME-MAJOR (T,B,·)
   const int H = w_hh.size(1);
   TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
   const int MT = ceil_div(B, BM);
@@ -245,17 +245,15 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   auto h0c = h0.contiguous();
   auto* h0p = reinterpret_cast<const __hip_bfloat16*>(h0c.data_ptr());
   for (int t = 0; t < T; ++t) {
-    const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * H;
-    const long h_rs = (t == 0) ? H : (long)T * H;
+    const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * B * H;
     const float* cp = (t == 0) ? c0.data_ptr<float>()
-                               : cs.data_ptr<float>() + (long)(t - 1) * H;
-    const long cp_rs = (t == 0) ? H : (long)T * H;
+                               : cs.data_ptr<float>() + (long)(t - 1) * B * H;
     hipLaunchKernelGGL(lstm_cell_fused, grid, dim3(THREADS), lds, stream(),
-        hp, h_rs, wp, xpp + (long)t * 4 * H, (long)T * 4 * H,
-        bias.data_ptr<float>(), cp, cp_rs,
-        hsp + (long)t * H, (long)T * H,
-        cs.data_ptr<float>() + (long)t * H, (long)T * H,
-        gp + (long)t * 4 * H, (long)T * 4 * H, B, H, MT);
+        hp, (long)H, wp, xpp + (long)t * B * 4 * H, (long)4 * H,
+        bias.data_ptr<float>(), cp, (long)H,
+        hsp + (long)t * B * H, (long)H,
+        cs.data_ptr<float>() + (long)t * B * H, (long)H,
+        gp + (long)t * B * 4 * H, (long)4 * H, B, H, MT);
   }
 }
 

@@ -81,50 +81,53 @@ static void launch_fwd_step(const at::Tensor& xp, const at::Tensor& bias,
                             long cp_off, long cp_rs, at::Tensor& hs,
                             at::Tensor& cs, at::Tensor& gates, int t, int B,
                             int T, int H) {
+  // TIME-MAJOR layout: xp/hs/cs/gates are (T, B, ·) contiguous, so slice t
+  // is a contiguous (B, ·) block — hipBLASLt sees contiguous operands and
+  // the cell kernel gets unit row strides.
   const int threads = 256;
   const int blocks = ceil_div((long)B * H, threads);
   hipLaunchKernelGGL((lstm_cell_fwd<ST>), dim3(blocks), dim3(threads), 0, stream(),
-      reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+      reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
       reinterpret_cast<const ST*>(rec.data_ptr()), (long)4 * H,
       bias.data_ptr<float>(),
       c_prev.data_ptr<float>() + cp_off, cp_rs,
-      reinterpret_cast<ST*>(hs.data_ptr()) + (long)t * H, (long)T * H,
-      cs.data_ptr<float>() + (long)t * H, (long)T * H,
-      reinterpret_cast<ST*>(gates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+      reinterpret_cast<ST*>(hs.data_ptr()) + (long)t * B * H, (long)H,
+      cs.data_ptr<float>() + (long)t * B * H, (long)H,
+      reinterpret_cast<ST*>(gates.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
       B, H);
 }
 
-// hs,cs,gates are (B,T,·) preallocated; xp (B,T,4H); h0 (B,H); c0 fp32 (B,H).
+// hs,cs,gates are (T,B,·) preallocated; xp (T,B,4H); h0 (B,H); c0 fp32 (B,H).
 void lstm_seq_forward_lib(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                           at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
                           at::Tensor cs, at::Tensor gates) {
   CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
   CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates);
-  const int B = xp.size(0), T = xp.size(1);
+  const int T = xp.size(0), B = xp.size(1);
   const int H = w_hh.size(1);
   auto w_hh_t = w_hh.t();
   auto rec = at::empty({B, 4 * H}, xp.options());
   CI_DISPATCH_FB(xp.scalar_type(), "lstm_seq_forward_lib", [&] {
     for (int t = 0; t < T; ++t) {
-      auto h_prev = (t == 0) ? h0 : hs.select(1, t - 1);
+      auto h_prev = (t == 0) ? h0 : hs.select(0, t - 1);
       at::mm_out(rec, h_prev, w_hh_t);
       if (t == 0) {
         launch_fwd_step<scalar_t>(xp, bias, rec, c0, 0, H, hs, cs, gates, t, B, T, H);
       } else {
-        launch_fwd_step<scalar_t>(xp, bias, rec, cs, (long)(t - 1) * H,
-                                  (long)T * H, hs, cs, gates, t, B, T, H);
+        launch_fwd_step<scalar_t>(xp, bias, rec, cs, (long)(t - 1) * B * H,
+                                  (long)H, hs, cs, gates, t, B, T, H);
       }
     }
   });
 }
 
-// reverse loop; dh0/dc0 are (B,H) fp32 outputs.
+// reverse loop over (T,B,·) time-major saves; dh0/dc0 are (B,H) fp32 outs.
 void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
                        at::Tensor gates, at::Tensor hs, at::Tensor cs,
                        at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
                        at::Tensor dh0, at::Tensor dc0) {
   CI_CHECK_CUDA(dhs); CI_CHECK_CONTIG(dhs); CI_CHECK_CONTIG(dgates);
-  const int B = dhs.size(0), T = dhs.size(1);
+  const int T = dhs.size(0), B = dhs.size(1);
   const int H = w_hh.size(1);
   auto dc_buf = dcT.clone();                       // (B,H) fp32 running dc
   auto dh_rec = dhT.contiguous();                  // (B,H) scalar running rec grad
@@ -138,19 +141,19 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
   CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
     for (int t = T - 1; t >= 0; --t) {
       const float* cprev = (t == 0) ? c0.data_ptr<float>()
-                                    : cs.data_ptr<float>() + (long)(t - 1) * H;
-      const long cprs = (t == 0) ? H : (long)T * H;
+                                    : cs.data_ptr<float>() + (long)(t - 1) * B * H;
       hipLaunchKernelGGL((lstm_cell_bwd<scalar_t>), dim3(blocks), dim3(threads), 0, stream(),
-          reinterpret_cast<const scalar_t*>(dhs.data_ptr()) + (long)t * H, (long)T * H,
+          reinterpret_cast<const scalar_t*>(dhs.data_ptr()) + (long)t * B * H, (long)H,
           reinterpret_cast<const scalar_t*>(dh_rec.data_ptr()), (long)H,
           dc_buf.data_ptr<float>(), (long)H,
-          reinterpret_cast<const scalar_t*>(gates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
-          cprev, cprs,
-          cs.data_ptr<float>() + (long)t * H, (long)T * H,
-          reinterpret_cast<scalar_t*>(dgates.data_ptr()) + (long)t * 4 * H, (long)T * 4 * H,
+          reinterpret_cast<const scalar_t*>(gates.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
+          cprev, (long)H,
+          cs.data_ptr<float>() + (long)t * B * H, (long)H,
+          reinterpret_cast<scalar_t*>(dgates.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
           B, H);
-      // dh_{t-1} recurrent contribution: dgates_t @ w_hh (NT layout)
-      at::mm_out(dh_rec, dgates.select(1, t), w_hh_nt);
+      // dh_{t-1} recurrent contribution: dgates_t @ w_hh (NT layout,
+      // contiguous A thanks to time-major dgates)
+      at::mm_out(dh_rec, dgates.select(0, t), w_hh_nt);
     }
   });
   dh0.copy_(dh_rec);

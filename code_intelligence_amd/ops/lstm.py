@@ -3,22 +3,29 @@
 Forward over a whole (B, T, In) sequence for ONE layer:
 
   * the input-side projection ``x @ W_ih^T + b`` has no time dependency and
-    runs as one large plain GEMM (hipBLASLt via torch.matmul — per the
-    MI355X design rules, library GEMMs are allowed for plain GEMMs);
+    runs as one large plain GEMM (hipBLASLt via torch.mm — per the MI355X
+    design rules, library GEMMs are allowed for plain GEMMs);
   * the recurrent part is the hot sequential loop: per timestep a
     (B,H)x(H,4H) GEMM + gate activations + c/h update. On ROCm this is a
     hand-written CDNA4 kernel — either the fully fused MFMA cell kernel
-    (lstm_gemm.hip, gates interleaved per hidden unit so the epilogue can
-    finish c/h locally) or a hipBLASLt GEMM + pointwise HIP kernel
-    (lstm_pointwise.hip), selected by CI_LSTM_MODE=fused|lib (default fused).
+    (lstm_gemm.hip) or a hipBLASLt GEMM + pointwise HIP kernel
+    (lstm_pointwise.hip), selected by CI_LSTM_MODE=fused|lib.
 
-Backward mirrors it: a sequential per-timestep pointwise+GEMM loop for
-dh/dc/dgates, then batched plain GEMMs for dW_ih, dW_hh, db, dx.
+Internally everything is TIME-MAJOR (T, B, ·): every per-step slice
+(h_prev, xp_t, dgates_t) is then a contiguous (B, ·) block, which is what
+both the hand-written kernels and hipBLASLt want (measured: strided
+batch-major A costs the backward dh GEMM ~40% — scripts/gemm_probe.py).
+The public API stays batch-first; the layer boundary transposes, and when
+the neighbouring layer also runs time-major the transposes cancel into
+no-op views.
+
+Backward mirrors forward: a sequential per-timestep pointwise+GEMM loop
+for dh/dc/dgates (NT layout via one weight transpose), then batched plain
+GEMMs for dW_ih, dW_hh, db, dx.
 
 Gate order follows PyTorch/cuDNN (i, f, g, o) for checkpoint parity.
-The CPU path is a straightforward PyTorch reference (tests compare the HIP
-kernels against it in fp32).
-"""
+The CPU path is a straightforward PyTorch reference (tests compare the
+HIP kernels against it in fp32)."""
 from __future__ import annotations
 
 import os
@@ -51,7 +58,7 @@ def _cpu_lstm_loop(x: Tensor, h0: Tensor, c0: Tensor, w_ih: Tensor, w_hh: Tensor
 
 
 class _FusedLSTMFunction(torch.autograd.Function):
-    """GPU path: HIP cell kernels, explicit backward."""
+    """GPU path: HIP cell kernels over time-major saves, explicit backward."""
 
     @staticmethod
     def forward(ctx, x, h0, c0, w_ih, w_hh, b_ih, b_hh):
@@ -59,48 +66,49 @@ class _FusedLSTMFunction(torch.autograd.Function):
         B, T, In = x.shape
         H = w_hh.shape[1]
         dt = x.dtype
-        # input projection: one plain GEMM over all timesteps
+        # time-major input; free when x is already a (T,B,·) transpose view
+        x_tm = x.transpose(0, 1).contiguous()
         bias = (b_ih + b_hh).to(torch.float32)
-        xp = torch.matmul(x.reshape(B * T, In), w_ih.t()).view(B, T, 4 * H)
-        hs = torch.empty(B, T, H, dtype=dt, device=x.device)
-        cs = torch.empty(B, T, H, dtype=torch.float32, device=x.device)
-        gates = torch.empty(B, T, 4 * H, dtype=dt, device=x.device)
+        xp = torch.mm(x_tm.view(T * B, In), w_ih.t()).view(T, B, 4 * H)
+        hs = torch.empty(T, B, H, dtype=dt, device=x.device)
+        cs = torch.empty(T, B, H, dtype=torch.float32, device=x.device)
+        gates = torch.empty(T, B, 4 * H, dtype=dt, device=x.device)
         mode = os.environ.get("CI_LSTM_MODE", "fused")
-        if mode == "fused":
+        if mode == "fused" and dt == torch.bfloat16:
             lib.lstm_seq_forward_fused(xp, bias, h0, c0.to(torch.float32), w_hh,
                                        hs, cs, gates)
         else:
             lib.lstm_seq_forward_lib(xp, bias, h0, c0.to(torch.float32), w_hh,
                                      hs, cs, gates)
-        ctx.save_for_backward(x, h0, c0, w_ih, w_hh, gates, hs, cs)
-        hT = hs[:, -1].clone()
-        cT = cs[:, -1].to(dt).clone()
-        return hs, hT, cT
+        ctx.save_for_backward(x_tm, h0, c0, w_ih, w_hh, gates, hs, cs)
+        hT = hs[-1].clone()
+        cT = cs[-1].to(dt)
+        out = hs.transpose(0, 1)  # (B,T,H) view of time-major storage
+        return out, hT, cT
 
     @staticmethod
     def backward(ctx, dhs, dhT, dcT):
         lib = ext.require()
-        x, h0, c0, w_ih, w_hh, gates, hs, cs = ctx.saved_tensors
-        B, T, In = x.shape
+        x_tm, h0, c0, w_ih, w_hh, gates, hs, cs = ctx.saved_tensors
+        T, B, In = x_tm.shape
         H = w_hh.shape[1]
-        dt = x.dtype
-        dgates = torch.empty(B, T, 4 * H, dtype=dt, device=x.device)
-        dh0 = torch.empty(B, H, dtype=torch.float32, device=x.device)
-        dc0 = torch.empty(B, H, dtype=torch.float32, device=x.device)
-        # sequential reverse loop: pointwise cell backward + per-step dh GEMM
-        lib.lstm_seq_backward(dhs.contiguous(), dhT.contiguous(),
+        dt = x_tm.dtype
+        dhs_tm = dhs.transpose(0, 1).contiguous()
+        dgates = torch.empty(T, B, 4 * H, dtype=dt, device=x_tm.device)
+        dh0 = torch.empty(B, H, dtype=torch.float32, device=x_tm.device)
+        dc0 = torch.empty(B, H, dtype=torch.float32, device=x_tm.device)
+        lib.lstm_seq_backward(dhs_tm, dhT.contiguous(),
                               dcT.to(torch.float32).contiguous(),
                               gates, hs, cs, c0.to(torch.float32), w_hh,
                               dgates, dh0, dc0)
-        dg2 = dgates.reshape(B * T, 4 * H)
-        # batched weight/input grads: plain GEMMs
-        dx = torch.matmul(dg2, w_ih).view(B, T, In)
-        dw_ih = torch.matmul(dg2.t(), x.reshape(B * T, In))
-        hprev = torch.cat([h0.unsqueeze(1), hs[:, :-1]], dim=1).reshape(B * T, H)
-        dw_hh = torch.matmul(dg2.t(), hprev)
+        dg2 = dgates.view(T * B, 4 * H)
+        dx_tm = torch.mm(dg2, w_ih).view(T, B, In)
+        dw_ih = torch.mm(dg2.t(), x_tm.view(T * B, In))
+        hprev = torch.cat([h0.unsqueeze(0), hs[:-1]], dim=0).view(T * B, H)
+        dw_hh = torch.mm(dg2.t(), hprev)
         db = dg2.sum(dim=0).to(dt)
-        return (dx, dh0.to(dt), dc0.to(dt), dw_ih.to(w_ih.dtype),
-                dw_hh.to(w_hh.dtype), db, db.clone())
+        return (dx_tm.transpose(0, 1), dh0.to(dt), dc0.to(dt),
+                dw_ih.to(w_ih.dtype), dw_hh.to(w_hh.dtype), db, db.clone())
 
 
 def lstm_forward(x: Tensor, h0: Tensor, c0: Tensor, w_ih: Tensor, w_hh: Tensor,
