@@ -1,36 +1,61 @@
 // K6 epilogues: fused tied-decoder softmax+CE row reductions.
 // The (chunked) logits GEMM runs on hipBLASLt; these kernels do the
 // per-row max/logsumexp pass and the in-place dlogits transform so full
-// logits never round-trip more than once (SURVEY.md §7 "hard parts": K6).
+// logits never round-trip more than once (SURVEY.md par.7 "hard parts").
+// The decoder bias is folded in HERE (template<HAS_BIAS>, vectorized
+// float4 bias loads — a per-element `bias ? bias[v] : 0` conditional made
+// hipcc branch around every load and cost 5x, the par.5 trap (c) of the
+// CDNA guide) so the (chunk, 60k) bias broadcast-add never materializes.
 #include "common.h"
 
 namespace ci {
 
-// one block per row: lse[i] = log(sum(exp(x - max))) + max ; tgt[i] = x[target]
-template <typename T, int THREADS>
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+template <typename T, int VEC>
+static __device__ __forceinline__ void load_vec(const T* p, float* out) {
+  T buf[VEC];
+  *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(p);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) out[e] = ld(buf + e);
+}
+
+template <int VEC>
+static __device__ __forceinline__ void load_bias(const float* b, float* out) {
+  #pragma unroll
+  for (int e = 0; e < VEC; e += 4)
+    *reinterpret_cast<f32x4*>(out + e) = *reinterpret_cast<const f32x4*>(b + e);
+}
+
+// one block per row: lse[i] = log(sum(exp(x + bias - max))) + max
+template <typename T, int THREADS, bool HAS_BIAS>
 __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride,
                                    const long* __restrict__ targets,
-                                   const float* __restrict__ bias,  // or null
+                                   const float* __restrict__ bias,
                                    float* __restrict__ lse,
                                    float* __restrict__ tgt, int V) {
   constexpr int NW = THREADS / kWave;
+  constexpr int VEC = 16 / sizeof(T);
   const int row = blockIdx.x;
   const T* x = logits + (long)row * row_stride;
   __shared__ float red[NW];
   __shared__ float bcast;
-  // pass 1: global row max (16-B vector loads, G13)
-  constexpr int VEC = 16 / sizeof(T);
   const int Vv = V / VEC * VEC;
+  float v8[VEC], b8[VEC];
+  // pass 1: global row max
   float mx = -3.4e38f;
   for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
-    T buf[VEC];
-    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
+    load_vec<T, VEC>(x + v, v8);
+    if (HAS_BIAS) {
+      load_bias<VEC>(bias + v, b8);
+      #pragma unroll
+      for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
+    }
     #pragma unroll
-    for (int e = 0; e < VEC; ++e)
-      mx = fmaxf(mx, ld(buf + e) + (bias ? bias[v + e] : 0.f));
+    for (int e = 0; e < VEC; ++e) mx = fmaxf(mx, v8[e]);
   }
   for (int v = Vv + threadIdx.x; v < V; v += THREADS)
-    mx = fmaxf(mx, ld(x + v) + (bias ? bias[v] : 0.f));
+    mx = fmaxf(mx, ld(x + v) + (HAS_BIAS ? bias[v] : 0.f));
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1)
     mx = fmaxf(mx, __shfl_down(mx, off));
@@ -44,17 +69,20 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
   __syncthreads();
   mx = bcast;
   __syncthreads();  // red[] reused below
-  // pass 2: sum exp(x - max)
+  // pass 2: sum exp
   float s = 0.f;
   for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
-    T buf[VEC];
-    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
+    load_vec<T, VEC>(x + v, v8);
+    if (HAS_BIAS) {
+      load_bias<VEC>(bias + v, b8);
+      #pragma unroll
+      for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
+    }
     #pragma unroll
-    for (int e = 0; e < VEC; ++e)
-      s += __expf(ld(buf + e) + (bias ? bias[v + e] : 0.f) - mx);
+    for (int e = 0; e < VEC; ++e) s += __expf(v8[e] - mx);
   }
   for (int v = Vv + threadIdx.x; v < V; v += THREADS)
-    s += __expf(ld(x + v) + (bias ? bias[v] : 0.f) - mx);
+    s += __expf(ld(x + v) + (HAS_BIAS ? bias[v] : 0.f) - mx);
   #pragma unroll
   for (int off = kWave / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
   if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = s;
@@ -63,8 +91,7 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
     float tot = 0.f;
     for (int w = 0; w < NW; ++w) tot += red[w];
     lse[row] = mx + __logf(tot);
-    tgt[row] = ld(x + targets[row]) +
-               (bias ? bias[targets[row]] : 0.f);
+    tgt[row] = ld(x + targets[row]) + (HAS_BIAS ? bias[targets[row]] : 0.f);
   }
 }
 
@@ -73,45 +100,57 @@ void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor bias,
   CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
   const int N = logits.size(0), V = logits.size(1);
   constexpr int THREADS = 256;
-  const float* bp = bias.numel() ? bias.data_ptr<float>() : nullptr;
+  const bool hb = bias.numel() > 0;
   CI_DISPATCH_FB(logits.scalar_type(), "ce_rowstats", [&] {
-    hipLaunchKernelGGL((ce_rowstats_kernel<scalar_t, THREADS>), dim3(N),
-        dim3(THREADS), 0, stream(),
-        reinterpret_cast<const scalar_t*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), bp, lse.data_ptr<float>(),
-        tgt.data_ptr<float>(), V);
+    if (hb) {
+      hipLaunchKernelGGL((ce_rowstats_kernel<scalar_t, THREADS, true>), dim3(N),
+          dim3(THREADS), 0, stream(),
+          reinterpret_cast<const scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), bias.data_ptr<float>(),
+          lse.data_ptr<float>(), tgt.data_ptr<float>(), V);
+    } else {
+      hipLaunchKernelGGL((ce_rowstats_kernel<scalar_t, THREADS, false>), dim3(N),
+          dim3(THREADS), 0, stream(),
+          reinterpret_cast<const scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), nullptr,
+          lse.data_ptr<float>(), tgt.data_ptr<float>(), V);
+    }
   });
 }
 
-// in-place: logits <- (exp(logits - lse) - onehot) * scale   (scale on device)
-// one block per row, vectorized 8-wide in-place transform (G13).
-template <typename T, int THREADS>
+// in-place: logits <- (exp(logits + bias - lse) - onehot) * scale
+template <typename T, int THREADS, bool HAS_BIAS>
 __global__ void ce_dlogits_kernel(T* __restrict__ logits, long row_stride,
                                   const long* __restrict__ targets,
-                                  const float* __restrict__ bias,  // or null
+                                  const float* __restrict__ bias,
                                   const float* __restrict__ lse,
-                                  const float* __restrict__ scale,
-                                  int V) {
+                                  const float* __restrict__ scale, int V) {
+  constexpr int VEC = 16 / sizeof(T);
   const int row = blockIdx.x;
   T* x = logits + (long)row * row_stride;
   const float l = lse[row];
   const float sc = scale[0];
   const int tgt = (int)targets[row];
-  constexpr int VEC = 16 / sizeof(T);  // one 16-B vector per thread-step
   const int Vv = V / VEC * VEC;
+  float v8[VEC], b8[VEC];
   for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    load_vec<T, VEC>(x + v, v8);
+    if (HAS_BIAS) {
+      load_bias<VEC>(bias + v, b8);
+      #pragma unroll
+      for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
+    }
     T buf[VEC];
-    *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(x + v);
     #pragma unroll
     for (int e = 0; e < VEC; ++e) {
-      float p = __expf(ld(buf + e) + (bias ? bias[v + e] : 0.f) - l);
+      float p = __expf(v8[e] - l);
       if (v + e == tgt) p -= 1.f;
       st(buf + e, p * sc);
     }
     *reinterpret_cast<int4*>(x + v) = *reinterpret_cast<const int4*>(buf);
   }
   for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
-    float p = __expf(ld(x + v) + (bias ? bias[v] : 0.f) - l);
+    float p = __expf(ld(x + v) + (HAS_BIAS ? bias[v] : 0.f) - l);
     if (v == tgt) p -= 1.f;
     st(x + v, p * sc);
   }
@@ -122,13 +161,21 @@ void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
   CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
   const int N = logits.size(0), V = logits.size(1);
   constexpr int THREADS = 256;
-  const float* bp = bias.numel() ? bias.data_ptr<float>() : nullptr;
+  const bool hb = bias.numel() > 0;
   CI_DISPATCH_FB(logits.scalar_type(), "ce_dlogits", [&] {
-    hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t, THREADS>),
-        dim3(N), dim3(THREADS), 0, stream(),
-        reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), bp, lse.data_ptr<float>(),
-        scale.data_ptr<float>(), V);
+    if (hb) {
+      hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t, THREADS, true>),
+          dim3(N), dim3(THREADS), 0, stream(),
+          reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), bias.data_ptr<float>(),
+          lse.data_ptr<float>(), scale.data_ptr<float>(), V);
+    } else {
+      hipLaunchKernelGGL((ce_dlogits_kernel<scalar_t, THREADS, false>),
+          dim3(N), dim3(THREADS), 0, stream(),
+          reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), nullptr,
+          lse.data_ptr<float>(), scale.data_ptr<float>(), V);
+    }
   });
 }
 
