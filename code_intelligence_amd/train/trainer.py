@@ -46,8 +46,12 @@ class LMTrainer:
         self.opt = FusedAdamW(self._param_groups(), lr=cfg.lr,
                               betas=cfg.betas, weight_decay=cfg.wd)
         self.lr_scale = 1.0
-        self.dist = DistributedGrads(model, bucket_mb=cfg.bucket_mb) \
-            if distributed else None
+        if distributed:
+            from ..parallel.ddp import broadcast_parameters
+            broadcast_parameters(model)
+            self.dist = DistributedGrads(model, bucket_mb=cfg.bucket_mb)
+        else:
+            self.dist = None
         self.global_step = 0
 
     def _param_groups(self):

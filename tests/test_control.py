@@ -1,0 +1,138 @@
+"""Control-plane tests: registry, ModelSync reconcile, needs-sync server,
+chatbot label matching (reference: Go test techniques — fakes, goldens)."""
+import json
+
+import pytest
+import yaml
+
+from code_intelligence_amd.control.chatbot import KubeflowLabels, create_app as chatbot_app
+from code_intelligence_amd.control.modelsync import ModelSync, ModelSyncSpec
+from code_intelligence_amd.control.needs_sync_server import (
+    create_app as sync_app, read_deployed_setter)
+from code_intelligence_amd.control.registry import LocalModelRegistry
+
+
+def test_registry_lifecycle(tmp_path):
+    reg = LocalModelRegistry(tmp_path)
+    rec = reg.create_training("labels")
+    assert reg.is_training("labels")
+    assert reg.latest_trained("labels") is None
+    reg.finish_training(rec.name, evaluation={"precision": 0.8, "recall": 0.6,
+                                              "confidence": 0.5})
+    assert not reg.is_training("labels")
+    assert reg.latest_trained("labels").name == rec.name
+    reg.deploy(rec.name)
+    assert reg.latest_deployed("labels").name == rec.name
+    # a newer deploy undeploys the old one
+    rec2 = reg.create_training("labels")
+    reg.finish_training(rec2.name)
+    reg.deploy(rec2.name)
+    assert reg.latest_deployed("labels").name == rec2.name
+    assert not reg.get(rec.name).deployed
+    assert reg.evaluation_at_confidence(rec.name)["precision"] == 0.8
+
+
+class FakeHTTP:
+    def __init__(self, payloads):
+        self.payloads = list(payloads)
+
+    def get(self, url, **kw):
+        class R:
+            def __init__(self, p):
+                self._p = p
+                self.status_code = 200
+
+            def json(self):
+                return self._p
+
+            def raise_for_status(self):
+                pass
+        return R(self.payloads.pop(0))
+
+
+def test_modelsync_reconcile_creates_and_skips_runs():
+    ran = []
+
+    def runner(cmd):
+        ran.append(cmd)
+        return 0
+
+    spec = ModelSyncSpec(name="sync", needs_sync_url="http://x/needsSync",
+                         run_command=["train"],
+                         parameter_mapping={"name": "model"})
+    ms = ModelSync(spec, session=FakeHTTP([
+        {"needsSync": True, "parameters": {"name": "m-123"}},
+        {"needsSync": True, "parameters": {"name": "m-123"}},
+        {"needsSync": False},
+    ]), runner=runner)
+    r1 = ms.reconcile()
+    assert "created" in r1
+    assert ran == [["train", "--model=m-123"]]  # parameter mapping applied
+    r2 = ms.reconcile()   # run Succeeded already (sync runner) -> creates again
+    assert "created" in r2
+    r3 = ms.reconcile()
+    assert r3["needs_sync"] is False
+    assert r3["succeeded"] == 2
+
+
+def test_modelsync_requeues_on_error():
+    class Boom:
+        def get(self, url, **kw):
+            raise ConnectionError("down")
+    spec = ModelSyncSpec(name="s", needs_sync_url="http://x", run_command=["t"])
+    ms = ModelSync(spec, session=Boom())
+    r = ms.reconcile()
+    assert r["requeue_after_s"] == spec.requeue_after_s
+
+
+def test_modelsync_history_gc():
+    spec = ModelSyncSpec(name="s", needs_sync_url="u", run_command=["t"],
+                         successful_runs_history_limit=2)
+    ms = ModelSync(spec, session=FakeHTTP(
+        [{"needsSync": True}] * 5 + [{"needsSync": False}]),
+        runner=lambda cmd: 0)
+    for _ in range(5):
+        ms.reconcile()
+    r = ms.reconcile()
+    assert r["succeeded"] == 2  # GC'd down to the history limit
+
+
+def test_needs_sync_server(tmp_path):
+    reg = LocalModelRegistry(tmp_path / "reg")
+    rec = reg.create_training("labels")
+    reg.finish_training(rec.name)
+    cfg = tmp_path / "Kptfile.yaml"
+    cfg.write_text(yaml.safe_dump({"openAPI": {"definitions": {
+        "io.k8s.cli.setters.automl-model": {
+            "x-k8s-cli": {"setter": {"name": "automl-model",
+                                     "value": "old-model"}}}}}}))
+    assert read_deployed_setter(cfg) == "old-model"
+    app = sync_app(reg, "labels", cfg)
+    c = app.test_client()
+    r = c.get("/needsSync").get_json()
+    assert r["needsSync"] is True
+    assert r["parameters"]["name"] == rec.name
+    # fresh model -> no retrain needed
+    assert c.get("/needsTrain").get_json()["needsTrain"] is False
+
+
+def test_chatbot_label_matching(tmp_path):
+    labels = KubeflowLabels([
+        {"name": "area/ops", "owners": ["alice"]},
+        {"name": "platform/gcp", "owners": ["bob", "carol"]},
+        {"name": "area/docs", "owners": []},
+    ])
+    # server.go table-driven matchLabels semantics
+    assert [l["name"] for l in labels.match_labels("area")] == \
+        ["area/ops", "area/docs"]
+    assert [l["name"] for l in labels.match_labels("platform", "gcp")] == \
+        ["platform/gcp"]
+    app = chatbot_app(labels)
+    c = app.test_client()
+    r = c.post("/dialogflow/webhook", json={
+        "queryResult": {"parameters": {"area": "platform", "value": "gcp"}}})
+    text = r.get_json()["fulfillmentText"]
+    assert "platform/gcp" in text and "bob" in text
+    r2 = c.post("/dialogflow/webhook", json={
+        "queryResult": {"parameters": {"area": "nosuch"}}})
+    assert "could not find" in r2.get_json()["fulfillmentText"]
