@@ -72,7 +72,10 @@ class RepoSpecificLabelModel(IssueLabelModel):
         if r.status_code != 200:
             log.warning("embedding service returned %s", r.status_code)
             return None
-        return np.frombuffer(r.content, dtype="<f4")[:EMBEDDING_DIM]
+        # keep the mean+max pools the classifier was trained on
+        # (1600 at the deployed config — embeddings.py:116)
+        keep = getattr(self.mlp, "in_dim", EMBEDDING_DIM)
+        return np.frombuffer(r.content, dtype="<f4")[:keep]
 
     def predict_issue_labels(self, org: str, repo: str, title: str,
                              text: List[str], context: Optional[dict] = None

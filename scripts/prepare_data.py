@@ -1,0 +1,83 @@
+"""Data pipeline: issue archive -> tokenized LM corpus
+(reference: Issue_Embeddings notebooks 01_AcquireData + 02_fastai_DataBunch:
+GHArchive -> mdparse pre-rules -> tokenize -> vocab -> databunch).
+
+Offline equivalent: reads JSONL archive shards (gh/bigquery.py format, or
+'synthetic:N' to generate), builds 'xxxfldtitle .. xxxfldbody ..' docs
+(process_dict), tokenizes with the framework rules in a process pool,
+builds a 60k vocab (min_freq 2), and writes docs.pt + vocab.json — the
+layout the train CLI consumes.
+
+  python scripts/prepare_data.py --archive /path/or/synthetic:5000 \
+      --out data_dir [--max_vocab 60000] [--workers 8]
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import multiprocessing as mp
+from pathlib import Path
+
+import torch
+
+from code_intelligence_amd.text.tokenizer import (Tokenizer, Vocab,
+                                                  process_dict)
+
+_tok = None
+
+
+def _init_worker():
+    global _tok
+    _tok = Tokenizer()
+
+
+def _tokenize(text: str):
+    return _tok.process_text(text)
+
+
+def load_archive_docs(archive: str) -> list[str]:
+    if archive.startswith("synthetic"):
+        from code_intelligence_amd.data.synthetic import synthetic_issue_texts
+        n = int(archive.split(":")[1]) if ":" in archive else 5000
+        raw = synthetic_issue_texts(n)
+    else:
+        raw = []
+        for f in sorted(Path(archive).glob("*.jsonl")):
+            for line in open(f):
+                if line.strip():
+                    ev = json.loads(line)
+                    raw.append({"title": ev.get("title", ""),
+                                "body": ev.get("body", "")})
+    return [process_dict(d)["text"] for d in raw]
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--archive", required=True)
+    p.add_argument("--out", required=True)
+    p.add_argument("--max_vocab", type=int, default=60000)
+    p.add_argument("--min_freq", type=int, default=2)
+    p.add_argument("--workers", type=int, default=mp.cpu_count())
+    args = p.parse_args()
+
+    texts = load_archive_docs(args.archive)
+    print(f"{len(texts)} documents")
+    if args.workers > 1:
+        with mp.Pool(args.workers, initializer=_init_worker) as pool:
+            token_docs = pool.map(_tokenize, texts, chunksize=64)
+    else:
+        _init_worker()
+        token_docs = [_tokenize(t) for t in texts]
+    vocab = Vocab.create(token_docs, max_vocab=args.max_vocab,
+                         min_freq=args.min_freq)
+    docs = [vocab.numericalize(t) for t in token_docs]
+    out = Path(args.out)
+    out.mkdir(parents=True, exist_ok=True)
+    torch.save(docs, out / "docs.pt")
+    vocab.save(out / "vocab.json")
+    n_tok = sum(len(d) for d in docs)
+    print(f"vocab {len(vocab)}; {n_tok} tokens -> {out}")
+
+
+if __name__ == "__main__":
+    main()

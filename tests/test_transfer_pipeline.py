@@ -1,0 +1,98 @@
+"""Transfer-learning pipeline tests: bulk embeddings, repo MLP training
+pipeline, TransferTrainer step (CPU)."""
+import numpy as np
+import pytest
+import torch
+
+from code_intelligence_amd.engine.embeddings import (CLASSIFIER_DIMS,
+                                                     get_all_issue_text)
+from code_intelligence_amd.engine.inference import InferenceWrapper
+from code_intelligence_amd.gh import bigquery
+from code_intelligence_amd.gh.gcs_util import ObjectStore
+from code_intelligence_amd.label.repo_specific_model import RepoSpecificLabelModel
+from code_intelligence_amd.label.trainers import (filter_labels, one_hot,
+                                                  run_training_pipeline)
+from code_intelligence_amd.models.awd_lstm import AWDLSTM
+from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+from code_intelligence_amd.train.transfer import TransferTrainer
+
+
+def _wrapper():
+    torch.manual_seed(0)
+    words = [f"w{i}" for i in range(200)]
+    vocab = Vocab(defaults_specials + words)
+    model = AWDLSTM(vocab_sz=len(vocab), emb_sz=16, n_hid=24, n_layers=2)
+    return InferenceWrapper(encoder=model.encoder, vocab=vocab, device="cpu")
+
+
+def _archive(tmp_path, n=80):
+    rng = np.random.default_rng(0)
+    events = []
+    for i in range(n):
+        labels = ["bug"] if i % 2 == 0 else ["feature"]
+        if i % 7 == 0:
+            labels.append("lifecycle/stale")  # must be excluded
+        events.append({"org": "o", "repo": "r", "issue_num": i,
+                       "title": f"w{i % 50} crash", "body": f"w{(i * 3) % 50} body",
+                       "labels": labels,
+                       "updated_at": "2024-01-01T00:00:00Z"})
+    bigquery.write_archive_events(events, tmp_path / "arch" / "s.jsonl")
+    return tmp_path / "arch"
+
+
+def test_filter_labels_and_one_hot():
+    lists = [["bug"]] * 35 + [["feature"]] * 31 + [["rare"]] * 5 + \
+        [["lifecycle/stale"]] * 40 + [["status/icebox"]] * 40
+    names = filter_labels(lists)
+    assert names == ["bug", "feature"]  # count>=30, lifecycle/status excluded
+    y = one_hot([["bug"], ["feature", "bug"], []], names)
+    assert y.tolist() == [[1, 0], [1, 1], [0, 0]]
+
+
+def test_bulk_embedding_truncates_to_classifier_dims(tmp_path):
+    w = _wrapper()
+    arch = _archive(tmp_path, n=10)
+    df, feats = get_all_issue_text("o", "r", w, archive_root=arch)
+    assert len(df) == 10
+    assert feats.shape[1] == min(CLASSIFIER_DIMS, 2 * 16)  # mean+max only
+
+
+def test_end_to_end_training_pipeline_serves(tmp_path):
+    """pipeline -> artifacts -> RepoSpecificLabelModel.from_repo loads."""
+    w = _wrapper()
+    arch = _archive(tmp_path, n=80)
+    store = ObjectStore(root=tmp_path / "store")
+    result = run_training_pipeline("o", "r", w, store=store, archive_root=arch)
+    assert set(result["labels"]) == {"bug", "feature"}
+
+    class _EmbSession:  # serve the wrapper's own embeddings over "HTTP"
+        def post(self, url, json=None, **kw):
+            class R:
+                pass
+            doc = w.process_dict({"title": json["title"], "body": json["body"]})
+            vec = w.get_pooled_features(doc["text"]).numpy().astype("<f4")
+            r = R()
+            r.status_code = 200
+            r.content = vec.tobytes()
+            return r
+
+    m = RepoSpecificLabelModel.from_repo("o", "r", store=store,
+                                         session=_EmbSession())
+    preds = m.predict_issue_labels("o", "r", "w2 crash", ["w6 body"])
+    assert isinstance(preds, dict)  # thresholds may or may not fire on tiny data
+
+
+def test_transfer_trainer_learns(tmp_path):
+    torch.manual_seed(0)
+    model = AWDLSTM(vocab_sz=100, emb_sz=16, n_hid=24, n_layers=2)
+    tr = TransferTrainer(model.encoder, n_labels=2, hidden=(16,), lr=5e-3)
+    ids = torch.randint(9, 100, (16, 12))
+    lens = torch.full((16,), 12)
+    # frozen encoder -> fixed features; the head fits 16 fixed labels.
+    # (a tiny random encoder yields weakly-separable features, so assert
+    # meaningful descent rather than full memorization — plumbing test)
+    y = (torch.rand(16, 2) < 0.5).float()
+    losses = [tr.train_step(ids, lens, y, lr=5e-2) for _ in range(400)]
+    assert losses[-1] < losses[0] * 0.8, (losses[0], losses[-1])
+    # encoder stayed frozen
+    assert all(not p.requires_grad for p in model.encoder.parameters())
