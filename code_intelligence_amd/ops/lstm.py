@@ -74,7 +74,7 @@ class _FusedLSTMFunction(torch.autograd.Function):
         cs = torch.empty(T, B, H, dtype=torch.float32, device=x.device)
         gates = torch.empty(T, B, 4 * H, dtype=dt, device=x.device)
         mode = os.environ.get("CI_LSTM_MODE", "fused")
-        if mode == "fused" and dt == torch.bfloat16:
+        if mode == "fused" and dt == torch.bfloat16 and H % 8 == 0:
             lib.lstm_seq_forward_fused(xp, bias, h0, c0.to(torch.float32), w_hh,
                                        hs, cs, gates)
         else:

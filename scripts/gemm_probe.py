@@ -4,6 +4,10 @@ Times hipBLASLt through torch.mm for the backward dh GEMM layouts and the
 CE chunk-size choices, so kernel work targets the measured best layout.
 Run: python scripts/gemm_probe.py
 """
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 import time
 
 import torch

@@ -11,6 +11,10 @@ layout the train CLI consumes.
   python scripts/prepare_data.py --archive /path/or/synthetic:5000 \
       --out data_dir [--max_vocab 60000] [--workers 8]
 """
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 from __future__ import annotations
 
 import argparse

@@ -7,6 +7,10 @@ per configuration.
 
 Run on an MI355X box: python scripts/serve_bench.py [--n 2000]
 """
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 from __future__ import annotations
 
 import argparse

@@ -4,6 +4,10 @@ encoder + MLP head, issues/sec per training step. DP-capable:
   python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
       --master-addr 127.0.0.1 scripts/transfer_bench.py
 """
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 from __future__ import annotations
 
 import argparse
