@@ -15,7 +15,6 @@ import sys
 from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
 
-from __future__ import annotations
 
 import argparse
 import json
