@@ -69,7 +69,9 @@ class InferenceWrapper:
         return self.vocab.numericalize(self.tokenizer.process_text(text))
 
     @staticmethod
-    def _bucket(n: int, buckets=(1, 2, 4, 8, 16, 32, 64, 128, 256, 512, 1024, 2048)) -> int:
+    def _bucket(n: int, buckets=(1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,
+                                 192, 256, 384, 512, 768, 1024, 1536, 2048)) -> int:
+        # ~1.5x-spaced buckets: worst-case padding 50%, typical ~20%
         for b in buckets:
             if n <= b:
                 return b
