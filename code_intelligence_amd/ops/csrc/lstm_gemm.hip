@@ -133,7 +133,7 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __hip_bfloat16* lsb = reinterpret_cast<__hip_bfloat16*>(smem);
-  // 3 buffers b: [A_b (128x64) | B_b (64x64)] — 1 tile spans each barrier
+  // buffer b: [A_b (128x64) | B_b (64x64)]
 #define LA(b) (lsb + (b) * (A_ELEMS + B_ELEMS))
 #define LB(b) (lsb + (b) * (A_ELEMS + B_ELEMS) + A_ELEMS)
 
@@ -143,14 +143,30 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
   const int nk = (K + BK - 1) / BK;
   const bool interior_rows = (row0_a + BM <= B) && (row0_b + BN <= N);
 
-  auto mfma_tile = [&](const __hip_bfloat16* la, const __hip_bfloat16* lb) {
+  auto stage = [&](int kt, int buf) {
+    const int k0 = kt * BK;
+    if (interior_rows && k0 + BK <= K) {
+      stage_glds<BM, false>(h_prev, h_rs, row0_a, k0, H, LA(buf));
+      stage_glds<BN, true>(w_hh, K, row0_b, k0, H, LB(buf));
+    } else {
+      stage_reg<BM, false>(h_prev, h_rs, row0_a, B, k0, K, H, LA(buf));
+      stage_reg<BN, true>(w_hh, K, row0_b, N, k0, K, H, LB(buf));
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int cur = kt & 1;
+    if (kt + 1 < nk) stage(kt + 1, cur ^ 1);  // async: flight hides under MFMA
     #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 af[4], bfr[2];
       #pragma unroll
-      for (int f = 0; f < 4; ++f) af[f] = frag(la, wm + f * 16, ks, lane);
+      for (int f = 0; f < 4; ++f) af[f] = frag(LA(cur), wm + f * 16, ks, lane);
       #pragma unroll
-      for (int f = 0; f < 2; ++f) bfr[f] = frag(lb, wn + f * 16, ks, lane);
+      for (int f = 0; f < 2; ++f) bfr[f] = frag(LB(cur), wn + f * 16, ks, lane);
       #pragma unroll
       for (int fm = 0; fm < 4; ++fm)
         #pragma unroll
@@ -158,61 +174,7 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
           acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
     }
-  };
-  constexpr int GLDS_PER_TILE = BM / 32 + BN / 32;  // per-wave glds per tile
-
-  const int nk_full = K / BK;  // glds-eligible full tiles
-  if (interior_rows && nk_full > 0) {
-    // --- pipelined interior path: 3 LDS buffers, counted vmcnt, raw
-    // barriers; one staged tile stays in flight ACROSS each barrier
-    // (cdna_hip_programming.md §5 'Pipelining across barriers').
-    stage_glds<BM, false>(h_prev, h_rs, row0_a, 0, H, LA(0));
-    stage_glds<BN, true>(w_hh, K, row0_b, 0, H, LB(0));
-    if (nk_full > 1) {
-      stage_glds<BM, false>(h_prev, h_rs, row0_a, BK, H, LA(1));
-      stage_glds<BN, true>(w_hh, K, row0_b, BK, H, LB(1));
-    }
-    for (int kt = 0; kt < nk_full; ++kt) {
-      // tile kt must be resident; tile kt+1 may stay in flight
-      if (kt + 1 < nk_full) {
-        asm volatile("s_waitcnt vmcnt(%0)" :: "n"(GLDS_PER_TILE) : "memory");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      // asm barrier WITH memory clobber: keeps the fragment ds_reads below
-      // the barrier (other waves' DMA portions of tile kt land only once
-      // every wave has passed its own vmcnt wait + this barrier)
-      asm volatile("s_barrier" ::: "memory");
-      if (kt + 2 < nk_full) {  // overwrites buf (kt+2)%3: reads done pre-barrier
-        const int b2 = (kt + 2) % 3;
-        stage_glds<BM, false>(h_prev, h_rs, row0_a, (kt + 2) * BK, H, LA(b2));
-        stage_glds<BN, true>(w_hh, K, row0_b, (kt + 2) * BK, H, LB(b2));
-      }
-      mfma_tile(LA(kt % 3), LB(kt % 3));
-    }
-    if (nk_full < nk) {  // ragged K tail via register staging
-      __syncthreads();
-      stage_reg<BM, false>(h_prev, h_rs, row0_a, B, nk_full * BK, K, H, LA(0));
-      stage_reg<BN, true>(w_hh, K, row0_b, N, nk_full * BK, K, H, LB(0));
-      __syncthreads();
-      mfma_tile(LA(0), LB(0));
-    }
-    __syncthreads();  // all LDS reads done before smem is reused below
-  } else {
-    // --- edge path (batch/N tails): register staging, 2 buffers
-    stage_reg<BM, false>(h_prev, h_rs, row0_a, B, 0, K, H, LA(0));
-    stage_reg<BN, true>(w_hh, K, row0_b, N, 0, K, H, LB(0));
-    __syncthreads();
-    for (int kt = 0; kt < nk; ++kt) {
-      const int cur = kt & 1;
-      if (kt + 1 < nk) {
-        stage_reg<BM, false>(h_prev, h_rs, row0_a, B, (kt + 1) * BK, K, H, LA(cur ^ 1));
-        stage_reg<BN, true>(w_hh, K, row0_b, N, (kt + 1) * BK, K, H, LB(cur ^ 1));
-      }
-      mfma_tile(LA(cur), LB(cur));
-      __syncthreads();
-    }
+    __syncthreads();  // drains the glds queue (vmcnt0) + barrier
   }
 
   // ---- epilogue: stash pre-activations in LDS, finish the cell ---------
@@ -274,7 +236,7 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   const int MT = ceil_div(B, BM);
   const dim3 grid(MT * ceil_div(4 * H, BN));
   const size_t lds = std::max(
-      (size_t)3 * (A_ELEMS + B_ELEMS) * sizeof(__hip_bfloat16),
+      (size_t)2 * (A_ELEMS + B_ELEMS) * sizeof(__hip_bfloat16),
       (size_t)BM * BN * sizeof(float));
   auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
   auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
