@@ -27,7 +27,9 @@ static __device__ __forceinline__ void load_bias(const float* b, float* out) {
     *reinterpret_cast<f32x4*>(out + e) = *reinterpret_cast<const f32x4*>(b + e);
 }
 
-// one block per row: lse[i] = log(sum(exp(x + bias - max))) + max
+// one block per row, ONLINE single-pass logsumexp (one read of the row
+// instead of two): per-thread running (max, sum) merged wave-wide then
+// across waves. lse[i] = log(sum(exp(x + bias - max))) + max.
 template <typename T, int THREADS, bool HAS_BIAS>
 __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride,
                                    const long* __restrict__ targets,
@@ -38,12 +40,11 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
   constexpr int VEC = 16 / sizeof(T);
   const int row = blockIdx.x;
   const T* x = logits + (long)row * row_stride;
-  __shared__ float red[NW];
-  __shared__ float bcast;
+  __shared__ float red_m[NW];
+  __shared__ float red_s[NW];
   const int Vv = V / VEC * VEC;
   float v8[VEC], b8[VEC];
-  // pass 1: global row max
-  float mx = -3.4e38f;
+  float m = -3.4e38f, sum = 0.f;
   for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
     load_vec<T, VEC>(x + v, v8);
     if (HAS_BIAS) {
@@ -51,46 +52,45 @@ __global__ void ce_rowstats_kernel(const T* __restrict__ logits, long row_stride
       #pragma unroll
       for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
     }
+    float bm = v8[0];
     #pragma unroll
-    for (int e = 0; e < VEC; ++e) mx = fmaxf(mx, v8[e]);
-  }
-  for (int v = Vv + threadIdx.x; v < V; v += THREADS)
-    mx = fmaxf(mx, ld(x + v) + (HAS_BIAS ? bias[v] : 0.f));
-  #pragma unroll
-  for (int off = kWave / 2; off > 0; off >>= 1)
-    mx = fmaxf(mx, __shfl_down(mx, off));
-  if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = mx;
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float m = red[0];
-    for (int w = 1; w < NW; ++w) m = fmaxf(m, red[w]);
-    bcast = m;
-  }
-  __syncthreads();
-  mx = bcast;
-  __syncthreads();  // red[] reused below
-  // pass 2: sum exp
-  float s = 0.f;
-  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
-    load_vec<T, VEC>(x + v, v8);
-    if (HAS_BIAS) {
-      load_bias<VEC>(bias + v, b8);
-      #pragma unroll
-      for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
+    for (int e = 1; e < VEC; ++e) bm = fmaxf(bm, v8[e]);
+    // rescale the running sum once per 8-wide block
+    if (bm > m) {
+      sum *= __expf(m - bm);
+      m = bm;
     }
     #pragma unroll
-    for (int e = 0; e < VEC; ++e) s += __expf(v8[e] - mx);
+    for (int e = 0; e < VEC; ++e) sum += __expf(v8[e] - m);
   }
-  for (int v = Vv + threadIdx.x; v < V; v += THREADS)
-    s += __expf(ld(x + v) + (HAS_BIAS ? bias[v] : 0.f) - mx);
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
+    const float val = ld(x + v) + (HAS_BIAS ? bias[v] : 0.f);
+    if (val > m) {
+      sum *= __expf(m - val);
+      m = val;
+    }
+    sum += __expf(val - m);
+  }
+  // wave merge
   #pragma unroll
-  for (int off = kWave / 2; off > 0; off >>= 1) s += __shfl_down(s, off);
-  if ((threadIdx.x & (kWave - 1)) == 0) red[threadIdx.x / kWave] = s;
+  for (int off = kWave / 2; off > 0; off >>= 1) {
+    const float om = __shfl_down(m, off);
+    const float os = __shfl_down(sum, off);
+    const float nm = fmaxf(m, om);
+    sum = sum * __expf(m - nm) + os * __expf(om - nm);
+    m = nm;
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0) {
+    red_m[threadIdx.x / kWave] = m;
+    red_s[threadIdx.x / kWave] = sum;
+  }
   __syncthreads();
   if (threadIdx.x == 0) {
-    float tot = 0.f;
-    for (int w = 0; w < NW; ++w) tot += red[w];
-    lse[row] = mx + __logf(tot);
+    float gm = red_m[0];
+    for (int w = 1; w < NW; ++w) gm = fmaxf(gm, red_m[w]);
+    float gs = 0.f;
+    for (int w = 0; w < NW; ++w) gs += red_s[w] * __expf(red_m[w] - gm);
+    lse[row] = gm + __logf(gs);
     tgt[row] = ld(x + targets[row]) + (HAS_BIAS ? bias[targets[row]] : 0.f);
   }
 }
