@@ -1,19 +1,26 @@
 """Fused tied-decoder softmax + cross-entropy (K6) — the FLOPs king.
 
 Reference semantics: fastai LinearDecoder + FlattenedLoss(CrossEntropy)
-over a 60k vocab (train.py:70, tie_weights/out_bias). Materializing
-(B*T, V) logits for the whole batch costs ~31 GB at the bench shape —
-instead the op chunks over rows (CHUNK=16384 measured fastest on MI355X,
-scripts/gemm_probe.py: 1.05 PF for the chunk GEMM): per chunk a plain
-hipBLASLt GEMM fills a preallocated logits tile, a HIP kernel reduces it
-to (logsumexp, target-logit) in one pass — the decoder BIAS is folded
-into that kernel so the (chunk, 60k) bias broadcast-add never
-materializes — and backward RECOMPUTES the chunk's logits, transforming
-them in place to dlogits = (softmax - onehot)/N. Only O(B*T) state (lse)
-is saved between forward and backward.
+over a 60k vocab (train.py:70, tie_weights/out_bias).
+
+MI355X design: chunked over rows (CHUNK=16384, ~1.05 PF for the chunk
+GEMM — scripts/gemm_probe.py): per chunk one plain hipBLASLt GEMM fills a
+logits tile and a HIP kernel reduces it to (logsumexp, target-logit) in a
+single online pass with the decoder bias folded in (the (chunk, 60k) bias
+broadcast-add never materializes).
+
+Backward: with 288 GB of HBM3E the full (B·T, 60k) bf16 logits (~31 GB at
+the bench shape) stay RESIDENT between forward and backward, so backward
+skips the logits recompute GEMM entirely and transforms the saved tile in
+place to dlogits = (softmax - onehot)/N (CE backward cost drops by one
+full 2.5e13-FLOP GEMM per step). If the allocation does not fit
+(CI_CE_SAVE_LOGITS=0 or OOM), it falls back to recompute-in-backward with
+only O(B·T) state saved.
 
 CPU path: plain F.cross_entropy composition (numerics reference)."""
 from __future__ import annotations
+
+import os
 
 import torch
 from torch import Tensor
@@ -45,15 +52,25 @@ class _FusedCEFunction(torch.autograd.Function):
         tgt_logit = torch.empty(N, dtype=torch.float32, device=h.device)
         b32 = bias.to(torch.float32) if bias is not None else _empty_f32(h.device)
         tgt64 = targets.to(torch.int64)
-        logits_buf = torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
+        save_logits = os.environ.get("CI_CE_SAVE_LOGITS", "1") != "0"
+        logits_full = None
+        if save_logits:
+            try:
+                logits_full = torch.empty(N, V, dtype=h.dtype, device=h.device)
+            except torch.cuda.OutOfMemoryError:
+                logits_full = None
+        scratch = None if logits_full is not None else \
+            torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
         w_t = weight.t()
         for s in range(0, N, C):
             e = min(N, s + C)
-            logits = logits_buf[: e - s]
+            logits = logits_full[s:e] if logits_full is not None \
+                else scratch[: e - s]
             torch.mm(h[s:e], w_t, out=logits)
             lib.ce_rowstats(logits, tgt64[s:e], b32, lse[s:e], tgt_logit[s:e])
         loss = (lse - tgt_logit).mean()
         ctx.save_for_backward(h, weight, b32, tgt64, lse)
+        ctx.logits_full = logits_full
         ctx.has_bias = bias is not None
         return loss
 
@@ -61,6 +78,8 @@ class _FusedCEFunction(torch.autograd.Function):
     def backward(ctx, dloss: Tensor):
         lib = ext.require()
         h, weight, b32, targets, lse = ctx.saved_tensors
+        logits_full = ctx.logits_full
+        ctx.logits_full = None  # release asap
         has_bias = ctx.has_bias
         N, H = h.shape
         V = weight.shape[0]
@@ -69,13 +88,17 @@ class _FusedCEFunction(torch.autograd.Function):
         dw = torch.zeros_like(weight, dtype=torch.float32)
         db = torch.zeros(V, dtype=torch.float32, device=h.device) if has_bias else None
         scale = (dloss / N).to(torch.float32).reshape(1)
-        logits_buf = torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
+        scratch = None if logits_full is not None else \
+            torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
         w_t = weight.t()
         bias_arg = b32 if has_bias else _empty_f32(h.device)
         for s in range(0, N, C):
             e = min(N, s + C)
-            dlog = logits_buf[: e - s]
-            torch.mm(h[s:e], w_t, out=dlog)
+            if logits_full is not None:
+                dlog = logits_full[s:e]
+            else:
+                dlog = scratch[: e - s]
+                torch.mm(h[s:e], w_t, out=dlog)  # recompute fallback
             # in-place: dlog <- (softmax(dlog + bias) - onehot) * scale
             lib.ce_dlogits(dlog, targets[s:e], bias_arg, lse[s:e], scale)
             torch.mm(dlog, weight, out=dh[s:e])
