@@ -40,14 +40,14 @@ def _empty_f32(device) -> Tensor:
 
 
 class _FusedCEFunction(torch.autograd.Function):
-    CHUNK = 16384
+    CHUNK = int(os.environ.get("CI_CE_CHUNK", "16384"))
 
     @staticmethod
     def forward(ctx, h: Tensor, weight: Tensor, bias: Tensor, targets: Tensor):
         lib = ext.require()
         N, H = h.shape
         V = weight.shape[0]
-        C = _FusedCEFunction.CHUNK
+        C = int(os.environ.get("CI_CE_CHUNK", _FusedCEFunction.CHUNK))
         lse = torch.empty(N, dtype=torch.float32, device=h.device)
         tgt_logit = torch.empty(N, dtype=torch.float32, device=h.device)
         b32 = bias.to(torch.float32) if bias is not None else _empty_f32(h.device)

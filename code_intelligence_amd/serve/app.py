@@ -34,7 +34,7 @@ def create_app(wrapper: InferenceWrapper | None = None,
     if wrapper is None:
         wrapper = InferenceWrapper(
             model_path=model_path or os.environ.get("MODEL_PATH", "model_files"),
-            use_graphs=os.environ.get("CI_SERVE_GRAPHS", "1") != "0")
+            use_graphs=os.environ.get("CI_SERVE_GRAPHS", "0") == "1")
     app.config["wrapper"] = wrapper
 
     @app.route("/healthz", methods=["GET"])
