@@ -7,7 +7,41 @@
 
 namespace ci {
 
-// one thread per (b, j): gates = xp + rec (+bias); c' = f*c + i*g; h' = o*tanh(c')
+// VEC consecutive j per thread with 16-B vector loads/stores (G13: scalar
+// bf16 loads cost ~2x; measured 9.9 us vs the ~5.7 us bandwidth floor).
+// Scalar tail path covers H % VEC != 0.
+template <typename T, int VEC>
+static __device__ __forceinline__ void ldv(const T* p, float* out) {
+  T buf[VEC];
+  *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(p);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) out[e] = ld(buf + e);
+}
+
+template <typename T, int VEC>
+static __device__ __forceinline__ void stv(T* p, const float* v) {
+  T buf[VEC];
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) st(buf + e, v[e]);
+  *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(buf);
+}
+
+template <int VEC>
+static __device__ __forceinline__ void ldv_f32(const float* p, float* out) {
+  #pragma unroll
+  for (int e = 0; e < VEC; e += 4)
+    *reinterpret_cast<float4*>(out + e) = *reinterpret_cast<const float4*>(p + e);
+}
+
+template <int VEC>
+static __device__ __forceinline__ void stv_f32(float* p, const float* v) {
+  #pragma unroll
+  for (int e = 0; e < VEC; e += 4)
+    *reinterpret_cast<float4*>(p + e) = *reinterpret_cast<const float4*>(v + e);
+}
+
+// one thread per (b, j-block): gates = xp + rec (+bias); c' = f*c + i*g;
+// h' = o*tanh(c')
 template <typename T>
 __global__ void lstm_cell_fwd(
     const T* __restrict__ xp, long xp_rs,      // (B,4H) view, row stride xp_rs
@@ -18,9 +52,53 @@ __global__ void lstm_cell_fwd(
     float* __restrict__ c_out, long c_rs,
     T* __restrict__ gates_out, long g_rs,      // post-activation i,f,g,o
     int B, int H) {
+  constexpr int VEC = 16 / sizeof(T);
+  const int Hv = H / VEC;  // vector blocks per row (tail handled scalar)
   const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= (long)B * H) return;
-  const int b = idx / H, j = idx % H;
+  if (idx >= (long)B * Hv) return;
+  const int b = idx / Hv, j = (idx % Hv) * VEC;
+  const long xo = (long)b * xp_rs + j;
+  const long ro = (long)b * rec_rs + j;
+  float gi[VEC], gf[VEC], gg[VEC], go[VEC], tmp[VEC], cprev[VEC];
+  ldv<T, VEC>(xp + xo, gi); ldv<T, VEC>(rec + ro, tmp);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) gi[e] = sigmoidf_(gi[e] + tmp[e] + bias[j + e]);
+  ldv<T, VEC>(xp + xo + H, gf); ldv<T, VEC>(rec + ro + H, tmp);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) gf[e] = sigmoidf_(gf[e] + tmp[e] + bias[j + H + e]);
+  ldv<T, VEC>(xp + xo + 2 * H, gg); ldv<T, VEC>(rec + ro + 2 * H, tmp);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) gg[e] = tanhf(gg[e] + tmp[e] + bias[j + 2 * H + e]);
+  ldv<T, VEC>(xp + xo + 3 * H, go); ldv<T, VEC>(rec + ro + 3 * H, tmp);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) go[e] = sigmoidf_(go[e] + tmp[e] + bias[j + 3 * H + e]);
+  ldv_f32<VEC>(c_prev + (long)b * cp_rs + j, cprev);
+  float c[VEC], h[VEC];
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) {
+    c[e] = gf[e] * cprev[e] + gi[e] * gg[e];
+    h[e] = go[e] * tanhf(c[e]);
+  }
+  stv<T, VEC>(h_out + (long)b * h_rs + j, h);
+  stv_f32<VEC>(c_out + (long)b * c_rs + j, c);
+  const long gout = (long)b * g_rs + j;
+  stv<T, VEC>(gates_out + gout, gi);
+  stv<T, VEC>(gates_out + gout + H, gf);
+  stv<T, VEC>(gates_out + gout + 2 * H, gg);
+  stv<T, VEC>(gates_out + gout + 3 * H, go);
+}
+
+// scalar tail kernel for H % VEC != 0 columns
+template <typename T>
+__global__ void lstm_cell_fwd_tail(
+    const T* __restrict__ xp, long xp_rs, const T* __restrict__ rec, long rec_rs,
+    const float* __restrict__ bias, const float* __restrict__ c_prev, long cp_rs,
+    T* __restrict__ h_out, long h_rs, float* __restrict__ c_out, long c_rs,
+    T* __restrict__ gates_out, long g_rs, int B, int H, int j0) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int ncol = H - j0;
+  if (idx >= (long)B * ncol) return;
+  const int b = idx / ncol, j = j0 + idx % ncol;
   const long xo = (long)b * xp_rs + j;
   const long ro = (long)b * rec_rs + j;
   float gi = ld(xp + xo) + ld(rec + ro) + bias[j];
@@ -39,9 +117,9 @@ __global__ void lstm_cell_fwd(
   st(gates_out + gout + 3 * H, go);
 }
 
-// backward pointwise: consumes dh_ext (from upstream) + dh_rec (from t+1 GEMM),
-// running dc (fp32 buffer, in/out), saved post-act gates, c_{t-1}, c_t.
-template <typename T>
+// backward pointwise: consumes dh_ext (upstream) + dh_rec (t+1 GEMM),
+// running dc (fp32, in/out), saved post-act gates, c_{t-1}, c_t.
+template <typename T, bool VECTOR>
 __global__ void lstm_cell_bwd(
     const T* __restrict__ dh_ext, long dhe_rs,
     const T* __restrict__ dh_rec, long dhr_rs,
@@ -51,28 +129,60 @@ __global__ void lstm_cell_bwd(
     const float* __restrict__ c_t, long ct_rs,
     T* __restrict__ dgates, long dg_rs,
     int B, int H) {
+  constexpr int VEC = VECTOR ? 16 / (int)sizeof(T) : 1;
+  const int Hv = H / VEC;
   const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (idx >= (long)B * H) return;
-  const int b = idx / H, j = idx % H;
-  const long go = (long)b * g_rs + j;
-  const float gi = ld(gates + go);
-  const float gf = ld(gates + go + H);
-  const float gg = ld(gates + go + 2 * H);
-  const float gout = ld(gates + go + 3 * H);
-  float dh = ld(dh_ext + (long)b * dhe_rs + j) + ld(dh_rec + (long)b * dhr_rs + j);
-  const float ct = c_t[(long)b * ct_rs + j];
-  const float tct = tanhf(ct);
-  const float do_ = dh * tct;
-  float dct = dc_buf[(long)b * dc_rs + j] + dh * gout * (1.f - tct * tct);
-  const float di = dct * gg;
-  const float df = dct * c_prev[(long)b * cp_rs + j];
-  const float dg = dct * gi;
-  dc_buf[(long)b * dc_rs + j] = dct * gf;  // dc_{t-1}
-  const long dgo = (long)b * dg_rs + j;
-  st(dgates + dgo, di * gi * (1.f - gi));
-  st(dgates + dgo + H, df * gf * (1.f - gf));
-  st(dgates + dgo + 2 * H, dg * (1.f - gg * gg));
-  st(dgates + dgo + 3 * H, do_ * gout * (1.f - gout));
+  if (idx >= (long)B * Hv) return;
+  const int b = idx / Hv, j = (idx % Hv) * VEC;
+  const long go_ = (long)b * g_rs + j;
+  float gi[VEC], gf[VEC], gg[VEC], gout[VEC], dh[VEC], tmp[VEC];
+  float ct[VEC], cprev[VEC], dcin[VEC];
+  if (VECTOR) {
+    ldv<T, VEC>(gates + go_, gi);
+    ldv<T, VEC>(gates + go_ + H, gf);
+    ldv<T, VEC>(gates + go_ + 2 * H, gg);
+    ldv<T, VEC>(gates + go_ + 3 * H, gout);
+    ldv<T, VEC>(dh_ext + (long)b * dhe_rs + j, dh);
+    ldv<T, VEC>(dh_rec + (long)b * dhr_rs + j, tmp);
+    ldv_f32<VEC>(c_t + (long)b * ct_rs + j, ct);
+    ldv_f32<VEC>(c_prev + (long)b * cp_rs + j, cprev);
+    ldv_f32<VEC>(dc_buf + (long)b * dc_rs + j, dcin);
+  } else {
+    gi[0] = ld(gates + go_); gf[0] = ld(gates + go_ + H);
+    gg[0] = ld(gates + go_ + 2 * H); gout[0] = ld(gates + go_ + 3 * H);
+    dh[0] = ld(dh_ext + (long)b * dhe_rs + j);
+    tmp[0] = ld(dh_rec + (long)b * dhr_rs + j);
+    ct[0] = c_t[(long)b * ct_rs + j];
+    cprev[0] = c_prev[(long)b * cp_rs + j];
+    dcin[0] = dc_buf[(long)b * dc_rs + j];
+  }
+  float dgi[VEC], dgf[VEC], dgg[VEC], dgo[VEC], dcout[VEC];
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) {
+    const float dhe = dh[e] + tmp[e];
+    const float tct = tanhf(ct[e]);
+    const float do_ = dhe * tct;
+    const float dct = dcin[e] + dhe * gout[e] * (1.f - tct * tct);
+    dgi[e] = dct * gg[e] * gi[e] * (1.f - gi[e]);
+    dgf[e] = dct * cprev[e] * gf[e] * (1.f - gf[e]);
+    dgg[e] = dct * gi[e] * (1.f - gg[e] * gg[e]);
+    dgo[e] = do_ * gout[e] * (1.f - gout[e]);
+    dcout[e] = dct * gf[e];
+  }
+  const long dgo_ = (long)b * dg_rs + j;
+  if (VECTOR) {
+    stv_f32<VEC>(dc_buf + (long)b * dc_rs + j, dcout);
+    stv<T, VEC>(dgates + dgo_, dgi);
+    stv<T, VEC>(dgates + dgo_ + H, dgf);
+    stv<T, VEC>(dgates + dgo_ + 2 * H, dgg);
+    stv<T, VEC>(dgates + dgo_ + 3 * H, dgo);
+  } else {
+    dc_buf[(long)b * dc_rs + j] = dcout[0];
+    st(dgates + dgo_, dgi[0]);
+    st(dgates + dgo_ + H, dgf[0]);
+    st(dgates + dgo_ + 2 * H, dgg[0]);
+    st(dgates + dgo_ + 3 * H, dgo[0]);
+  }
 }
 
 template <typename ST>
@@ -85,16 +195,28 @@ static void launch_fwd_step(const at::Tensor& xp, const at::Tensor& bias,
   // is a contiguous (B, ·) block — hipBLASLt sees contiguous operands and
   // the cell kernel gets unit row strides.
   const int threads = 256;
-  const int blocks = ceil_div((long)B * H, threads);
-  hipLaunchKernelGGL((lstm_cell_fwd<ST>), dim3(blocks), dim3(threads), 0, stream(),
-      reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
-      reinterpret_cast<const ST*>(rec.data_ptr()), (long)4 * H,
-      bias.data_ptr<float>(),
-      c_prev.data_ptr<float>() + cp_off, cp_rs,
-      reinterpret_cast<ST*>(hs.data_ptr()) + (long)t * B * H, (long)H,
-      cs.data_ptr<float>() + (long)t * B * H, (long)H,
-      reinterpret_cast<ST*>(gates.data_ptr()) + (long)t * B * 4 * H, (long)4 * H,
-      B, H);
+  constexpr int VEC = 16 / sizeof(ST);
+  const int Hv = H / VEC;
+  const ST* xpp = reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * B * 4 * H;
+  const ST* recp = reinterpret_cast<const ST*>(rec.data_ptr());
+  ST* hp = reinterpret_cast<ST*>(hs.data_ptr()) + (long)t * B * H;
+  float* cp = cs.data_ptr<float>() + (long)t * B * H;
+  ST* gp = reinterpret_cast<ST*>(gates.data_ptr()) + (long)t * B * 4 * H;
+  if (Hv > 0) {
+    hipLaunchKernelGGL((lstm_cell_fwd<ST>),
+        dim3(ceil_div((long)B * Hv, threads)), dim3(threads), 0, stream(),
+        xpp, (long)4 * H, recp, (long)4 * H, bias.data_ptr<float>(),
+        c_prev.data_ptr<float>() + cp_off, cp_rs,
+        hp, (long)H, cp, (long)H, gp, (long)4 * H, B, H);
+  }
+  if (H % VEC) {
+    const int j0 = Hv * VEC;
+    hipLaunchKernelGGL((lstm_cell_fwd_tail<ST>),
+        dim3(ceil_div((long)B * (H - j0), threads)), dim3(threads), 0, stream(),
+        xpp, (long)4 * H, recp, (long)4 * H, bias.data_ptr<float>(),
+        c_prev.data_ptr<float>() + cp_off, cp_rs,
+        hp, (long)H, cp, (long)H, gp, (long)4 * H, B, H, j0);
+  }
 }
 
 // hs,cs,gates are (T,B,·) preallocated; xp (T,B,4H); h0 (B,H); c0 fp32 (B,H).
@@ -137,12 +259,16 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
   auto w_hh_tc = w_hh.t().contiguous();
   auto w_hh_nt = w_hh_tc.t();
   const int threads = 256;
-  const int blocks = ceil_div((long)B * H, threads);
   CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    const bool vec_ok = (H % VEC) == 0;
+    const int blocks = ceil_div((long)B * (vec_ok ? H / VEC : H), threads);
     for (int t = T - 1; t >= 0; --t) {
       const float* cprev = (t == 0) ? c0.data_ptr<float>()
                                     : cs.data_ptr<float>() + (long)(t - 1) * B * H;
-      hipLaunchKernelGGL((lstm_cell_bwd<scalar_t>), dim3(blocks), dim3(threads), 0, stream(),
+      auto kern = vec_ok ? lstm_cell_bwd<scalar_t, true>
+                         : lstm_cell_bwd<scalar_t, false>;
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(threads), 0, stream(),
           reinterpret_cast<const scalar_t*>(dhs.data_ptr()) + (long)t * B * H, (long)H,
           reinterpret_cast<const scalar_t*>(dh_rec.data_ptr()), (long)H,
           dc_buf.data_ptr<float>(), (long)H,
