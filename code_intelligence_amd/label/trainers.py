@@ -115,3 +115,63 @@ def run_training_pipeline(org: str, repo: str, wrapper: InferenceWrapper,
     """Both steps (Training_Pipeline.ipynb @dsl.pipeline equivalent)."""
     save_issue_embeddings(org, repo, wrapper, store, archive_root)
     return train_repo_mlp(org, repo, store, device=device)
+
+
+# --- universal kind model training (reference: the Keras universal model
+# trained on kind labels; here the torch CNN of universal_kind_label_model) --
+def kind_targets(label_lists: List[List[str]]) -> np.ndarray:
+    """3-class targets (bug/feature/question) from raw labels; accepts both
+    bare and 'kind/'-prefixed forms."""
+    classes = ("bug", "feature", "question")
+    y = np.zeros((len(label_lists), 3), dtype=np.float32)
+    for r, labels in enumerate(label_lists):
+        for l in labels:
+            name = l.split("/", 1)[-1].lower()
+            if name in classes:
+                y[r, classes.index(name)] = 1.0
+    return y
+
+
+def train_universal_model(org: str, archive_root=None, epochs: int = 5,
+                          lr: float = 1e-3, max_vocab: int = 20000,
+                          device: str = "cpu", batch_size: int = 64,
+                          prefix: str = ""):
+    """Train the universal 3-kind classifier from the issue archive.
+    Returns a ready UniversalKindLabelModel."""
+    import torch
+    from ..gh import bigquery
+    from ..text.tokenizer import Tokenizer, Vocab
+    from .universal_kind_label_model import (UniversalKindLabelModel,
+                                             UniversalKindNet)
+
+    df = bigquery.get_issues(org, archive_root=archive_root)
+    if df.empty:
+        raise ValueError(f"no archived issues for org {org}")
+    tok = Tokenizer()
+    docs = [tok.process_text(f"{t}\n{b}")[:256]
+            for t, b in zip(df["title"], df["body"])]
+    vocab = Vocab.create(docs, max_vocab=max_vocab, min_freq=1)
+    y = torch.tensor(kind_targets(df["labels"].tolist()))
+    ids = [torch.tensor(vocab.numericalize(d) or [0]) for d in docs]
+    maxlen = max(len(i) for i in ids)
+    X = torch.full((len(ids), maxlen), 1, dtype=torch.int64)  # pad=1
+    for r, seq in enumerate(ids):
+        X[r, :len(seq)] = seq
+    net = UniversalKindNet(len(vocab)).to(device)
+    opt = torch.optim.AdamW(net.parameters(), lr=lr)
+    lossf = torch.nn.BCEWithLogitsLoss()
+    net.train()
+    X, y = X.to(device), y.to(device)
+    for _ in range(epochs):
+        perm = torch.randperm(len(X), device=device)
+        for s in range(0, len(X), batch_size):
+            b = perm[s:s + batch_size]
+            opt.zero_grad()
+            loss = lossf(net(X[b]), y[b])
+            loss.backward()
+            opt.step()
+    net.eval()
+    model = UniversalKindLabelModel(net.cpu(), vocab, prefix=prefix)
+    log.info("trained universal model on %d issues (final loss %.4f)",
+             len(X), float(loss))
+    return model
