@@ -1,0 +1,70 @@
+"""Isolated kernel workload for PMC counter capture (rocprofv3 --pmc).
+
+Runs each hand-written gfx950 kernel repeatedly at the deployed shapes so
+per-dispatch counters (MfmaUtil, LDS bank conflicts, occupancy, ...) can
+be aggregated per kernel. Keep the workload small: PMC collection
+serializes dispatches.
+
+  cd /tmp && export TMPDIR=/tmp
+  rocprofv3 --pmc MfmaUtil VALUBusy SQ_LDS_BANK_CONFLICT OccupancyPercent \
+      --output-format csv -d OUT -o pmc -- python scripts/kernel_pmc.py
+"""
+import sys
+from pathlib import Path
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
+import os
+
+import torch
+
+from code_intelligence_amd.ops import extension
+
+assert torch.cuda.is_available()
+lib = extension.require()
+dev = "cuda:0"
+torch.manual_seed(0)
+
+B, H, T = 512, 2400, 8
+dt = torch.bfloat16
+
+# fused MFMA LSTM cell kernel (K2)
+xp = torch.randn(T, B, 4 * H, device=dev, dtype=dt)
+bias = torch.randn(4 * H, device=dev, dtype=torch.float32)
+h0 = torch.randn(B, H, device=dev, dtype=dt) * 0.1
+c0 = torch.randn(B, H, device=dev, dtype=torch.float32) * 0.1
+w = (torch.randn(4 * H, H, device=dev) * 0.02).to(dt)
+hs = torch.empty(T, B, H, device=dev, dtype=dt)
+cs = torch.empty(T, B, H, device=dev, dtype=torch.float32)
+gates = torch.empty(T, B, 4 * H, device=dev, dtype=dt)
+lib.lstm_seq_forward_fused(xp, bias, h0, c0, w, hs, cs, gates)
+
+# pointwise cell fwd/bwd (lib mode)
+lib.lstm_seq_forward_lib(xp, bias, h0, c0, w, hs, cs, gates)
+dhs = torch.randn_like(hs)
+dgates = torch.empty_like(gates)
+dh0 = torch.empty(B, H, device=dev, dtype=torch.float32)
+dc0 = torch.empty(B, H, device=dev, dtype=torch.float32)
+lib.lstm_seq_backward(dhs, dhs[-1].clone(), c0.clone(), gates, hs, cs, c0, w,
+                      dgates, dh0, dc0)
+
+# CE epilogues (K6) at chunk shape
+N, V = 16384, 60000
+logits = torch.randn(N, V, device=dev, dtype=dt)
+tgt = torch.randint(0, V, (N,), device=dev)
+b32 = torch.randn(V, device=dev, dtype=torch.float32)
+lse = torch.empty(N, device=dev, dtype=torch.float32)
+tl = torch.empty(N, device=dev, dtype=torch.float32)
+for _ in range(3):
+    lib.ce_rowstats(logits, tgt, b32, lse, tl)
+scale = torch.full((1,), 1e-5, device=dev)
+for _ in range(3):
+    lib.ce_dlogits(logits, tgt, b32, lse, scale)
+
+# concat-pool (K5) at serve shape
+hid = torch.randn(200, 512, 800, device=dev, dtype=dt)
+lens = torch.randint(1, 513, (200,), device=dev, dtype=torch.int32)
+for _ in range(3):
+    lib.concat_pool(hid, lens)
+
+torch.cuda.synchronize()
+print("pmc workload done")
