@@ -72,11 +72,14 @@ class LMTrainer:
                                y.reshape(-1))
         if self.model.training:
             if self.cfg.alpha:  # activation regularization on dropped output
-                loss = loss + self.cfg.alpha * out.float().pow(2).mean()
+                # bf16 squares with fp32 accumulation: avoids materializing
+                # an fp32 copy of the (B,T,H) activations
+                loss = loss + self.cfg.alpha * out.pow(2).mean(dtype=torch.float32)
             if self.cfg.beta:   # temporal AR on raw output diffs
-                r = raw_outputs[-1].float()
+                r = raw_outputs[-1]
                 if r.shape[1] > 1:
-                    loss = loss + self.cfg.beta * (r[:, 1:] - r[:, :-1]).pow(2).mean()
+                    loss = loss + self.cfg.beta * (
+                        (r[:, 1:] - r[:, :-1]).pow(2).mean(dtype=torch.float32))
         return loss
 
     def train_step(self, x, y, lr: float, mom: Optional[float] = None) -> float:
