@@ -33,10 +33,8 @@ namespace ci {
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-constexpr int BM = 128, BN = 64, BK = 64;
-constexpr int THREADS = 256;             // 4 waves: 2(M) x 2(N) of 64x32
-constexpr int A_ELEMS = BM * BK;         // bf16 elements per A tile
-constexpr int B_ELEMS = BN * BK;
+constexpr int BK = 64;
+constexpr int THREADS = 256;             // 4 waves: 2(M) x 2(N)
 
 // element offset of (row, col) in the swizzled [rows][BK] bf16 LDS tile:
 // 16B chunks within a 128B row are XOR'd by (row&7) (T2).
@@ -106,6 +104,7 @@ static __device__ __forceinline__ bf16x8 frag(const __hip_bfloat16* lds,
   return *reinterpret_cast<const bf16x8*>(lds + swz(row, col));
 }
 
+template <int BM, int BN>
 __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
     const __hip_bfloat16* __restrict__ h_prev, long h_rs,
     const __hip_bfloat16* __restrict__ w_hh,   // (4H, H) checkpoint layout
@@ -128,16 +127,20 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
   const int N = 4 * H, K = H;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = (wave & 1) * 64;    // wave row offset (2 waves over 128)
-  const int wn = (wave >> 1) * 32;   // wave col offset (2 waves over 64)
+  const int wm = (wave & 1) * (BM / 2);   // wave row offset
+  const int wn = (wave >> 1) * (BN / 2);   // wave col offset
 
+  constexpr int A_ELEMS = BM * BK;       // bf16 elements per A tile
+  constexpr int B_ELEMS = BN * BK;
+  constexpr int FM = BM / 2 / 16;        // 16x16 fragments per wave (M)
+  constexpr int FN = BN / 2 / 16;        //                       (N)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __hip_bfloat16* lsb = reinterpret_cast<__hip_bfloat16*>(smem);
-  // buffer b: [A_b (128x64) | B_b (64x64)]
+  // buffer b: [A_b (BMx64) | B_b (BNx64)]
 #define LA(b) (lsb + (b) * (A_ELEMS + B_ELEMS))
 #define LB(b) (lsb + (b) * (A_ELEMS + B_ELEMS) + A_ELEMS)
 
-  f32x4 acc[4][2] = {};
+  f32x4 acc[FM][FN] = {};
 
   const int row0_a = mt * BM, row0_b = nt * BN;
   const int nk = (K + BK - 1) / BK;
@@ -178,12 +181,12 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
   }
 
   // ---- epilogue: stash pre-activations in LDS, finish the cell ---------
-  float* pre = reinterpret_cast<float*>(smem);   // [BM][BN] fp32 = 32 KiB
+  float* pre = reinterpret_cast<float*>(smem);   // [BM][BN] fp32
   constexpr int PRS = BN;
   #pragma unroll
-  for (int fm = 0; fm < 4; ++fm) {
+  for (int fm = 0; fm < FM; ++fm) {
     #pragma unroll
-    for (int fn = 0; fn < 2; ++fn) {
+    for (int fn = 0; fn < FN; ++fn) {
       const int col = wn + fn * 16 + (lane & 15);
       #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
@@ -194,11 +197,12 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
   }
   __syncthreads();
 
-  // BN=64 cols = 16 hidden units; thread p2 handles (b, j) pairs
+  // BN cols = BN/4 hidden units; thread p2 handles (b, j) pairs
   const int jt = nt * (BN / 4);
-  for (int p2 = threadIdx.x; p2 < BM * (BN / 4); p2 += THREADS) {
-    const int br = p2 >> 4;           // row within tile (BN/4 = 16)
-    const int jj = p2 & 15;           // unit within tile
+  constexpr int JPT = BN / 4;
+  for (int p2 = threadIdx.x; p2 < BM * JPT; p2 += THREADS) {
+    const int br = p2 / JPT;          // row within tile
+    const int jj = p2 % JPT;          // unit within tile
     const int b = row0_a + br;
     const int j = jt + jj;
     if (b >= B || j >= H) continue;
@@ -222,21 +226,16 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
 }
 
 // driver: whole-sequence forward, one fused launch per timestep.
-void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
-                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
-                            at::Tensor cs, at::Tensor gates) {
-  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
-  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates);
-  TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16,
-              "fused LSTM cell kernel is bf16; use CI_LSTM_MODE=lib for fp32");
-  TORCH_CHECK(w_hh.is_contiguous(), "w_hh must be contiguous");
+template <int BM, int BN>
+static void run_fused(at::Tensor& xp, at::Tensor& bias, at::Tensor& h0,
+                      at::Tensor& c0, at::Tensor& w_hh, at::Tensor& hs,
+                      at::Tensor& cs, at::Tensor& gates) {
   const int T = xp.size(0), B = xp.size(1);   // TIME-MAJOR (T,B,·)
   const int H = w_hh.size(1);
-  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
   const int MT = ceil_div(B, BM);
   const dim3 grid(MT * ceil_div(4 * H, BN));
   const size_t lds = std::max(
-      (size_t)2 * (A_ELEMS + B_ELEMS) * sizeof(__hip_bfloat16),
+      (size_t)2 * (BM + BN) * BK * sizeof(__hip_bfloat16),
       (size_t)BM * BN * sizeof(float));
   auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
   auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
@@ -248,12 +247,30 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
     const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * B * H;
     const float* cp = (t == 0) ? c0.data_ptr<float>()
                                : cs.data_ptr<float>() + (long)(t - 1) * B * H;
-    hipLaunchKernelGGL(lstm_cell_fused, grid, dim3(THREADS), lds, stream(),
+    hipLaunchKernelGGL((lstm_cell_fused<BM, BN>), grid, dim3(THREADS), lds,
+        stream(),
         hp, (long)H, wp, xpp + (long)t * B * 4 * H, (long)4 * H,
         bias.data_ptr<float>(), cp, (long)H,
         hsp + (long)t * B * H, (long)H,
         cs.data_ptr<float>() + (long)t * B * H, (long)H,
         gp + (long)t * B * 4 * H, (long)4 * H, B, H, MT);
+  }
+}
+
+void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
+                            at::Tensor cs, at::Tensor gates) {
+  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
+  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates);
+  TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16,
+              "fused LSTM cell kernel is bf16; use CI_LSTM_MODE=lib for fp32");
+  TORCH_CHECK(w_hh.is_contiguous(), "w_hh must be contiguous");
+  TORCH_CHECK(w_hh.size(1) % 8 == 0, "H must be a multiple of 8");
+  const char* tile = getenv("CI_FUSED_TILE");
+  if (tile && std::string(tile) == "64x64") {
+    run_fused<64, 64>(xp, bias, h0, c0, w_hh, hs, cs, gates);
+  } else {
+    run_fused<128, 64>(xp, bias, h0, c0, w_hh, hs, cs, gates);
   }
 }
 
