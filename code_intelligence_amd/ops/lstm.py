@@ -108,8 +108,12 @@ class _FusedLSTMFunction(torch.autograd.Function):
         dg2 = dgates.view(T * B, 4 * H)
         dx_tm = torch.mm(dg2, w_ih).view(T, B, In)
         dw_ih = torch.mm(dg2.t(), x_tm.view(T * B, In))
-        hprev = torch.cat([h0.unsqueeze(0), hs[:-1]], dim=0).view(T * B, H)
-        dw_hh = torch.mm(dg2.t(), hprev)
+        # dW_hh = sum_t h_{t-1}^T dgates_t — time-major slices stay
+        # contiguous, so no (T,B,H) cat materializes:
+        dw_hh = torch.mm(dgates[0].t(), h0.to(dt))
+        if T > 1:
+            dw_hh += torch.mm(dgates[1:].reshape((T - 1) * B, 4 * H).t(),
+                              hs[:-1].reshape((T - 1) * B, H))
         db = dg2.sum(dim=0).to(dt)
         return (dx_tm.transpose(0, 1), dh0.to(dt), dc0.to(dt),
                 dw_ih.to(w_ih.dtype), dw_hh.to(w_hh.dtype), db, db.clone())

@@ -48,7 +48,8 @@ def build_argparser() -> argparse.ArgumentParser:
 def load_docs(data_path: str, vocab_sz: int):
     if data_path.startswith("synthetic"):
         n = int(data_path.split(":")[1]) if ":" in data_path else 2000
-        docs = synthetic_issue_tokens(n, vocab_sz)
+        docs = synthetic_issue_tokens(n, vocab_sz,
+                                      markov="markov" in data_path)
         return docs, vocab_sz
     root = Path(data_path)
     docs = torch.load(root / "docs.pt", weights_only=True)
