@@ -79,6 +79,35 @@ class GitHubApp:
         r.raise_for_status()
         return r.json()  # {'token': ..., 'expires_at': ISO8601}
 
+    def list_installation_repos(self, installation_id: int) -> list:
+        """Repositories an installation can access (github_app.py parity)."""
+        token = self.get_installation_access_token(installation_id)["token"]
+        repos, page = [], 1
+        while True:
+            r = self.session.get(
+                f"{self.api_url}/installation/repositories",
+                params={"per_page": 100, "page": page},
+                headers={"Authorization": f"token {token}",
+                         "Accept": "application/vnd.github.v3+json"})
+            r.raise_for_status()
+            batch = r.json().get("repositories", [])
+            repos.extend(batch)
+            if len(batch) < 100:
+                return repos
+            page += 1
+
+    def add_reaction(self, owner: str, repo: str, comment_id: int,
+                     content: str, token: str) -> dict:
+        """React to an issue comment (worker feedback loop parity)."""
+        r = self.session.post(
+            f"{self.api_url}/repos/{owner}/{repo}/issues/comments/"
+            f"{comment_id}/reactions",
+            json={"content": content},
+            headers={"Authorization": f"token {token}",
+                     "Accept": "application/vnd.github.squirrel-girl-preview+json"})
+        r.raise_for_status()
+        return r.json()
+
 
 class FixedAccessTokenGenerator:
     """A constant personal-access-token source (github_app.py:265-303)."""

@@ -193,3 +193,56 @@ def test_bigquery_archive_dedupe(tmp_path):
     assert len(df) == 1
     assert df.iloc[0]["title"] == "new"
     assert df.iloc[0]["labels"] == ["bug", "feature"]
+
+
+def test_es256_against_openssl(tmp_path):
+    from code_intelligence_amd.gh.es256 import (chatbot_test_jwt,
+                                                parse_ec_private_key_pem,
+                                                sign_es256)
+    key = tmp_path / "ec.pem"
+    subprocess.run(["openssl", "ecparam", "-genkey", "-name", "prime256v1",
+                    "-noout", "-out", str(key)], check=True, capture_output=True)
+    pub = tmp_path / "ecpub.pem"
+    subprocess.run(["openssl", "ec", "-in", str(key), "-pubout", "-out", str(pub)],
+                   check=True, capture_output=True)
+    d = parse_ec_private_key_pem(key.read_text())
+    msg = b"chatbot webhook test"
+    r, s = sign_es256(msg, d)
+    # DER-encode (r, s) for openssl verification
+    def _int(v):
+        b = v.to_bytes(32, "big").lstrip(b"\x00")
+        if b[0] & 0x80:
+            b = b"\x00" + b
+        return b"\x02" + bytes([len(b)]) + b
+    body = _int(r) + _int(s)
+    der = b"\x30" + bytes([len(body)]) + body
+    (tmp_path / "m").write_bytes(msg)
+    (tmp_path / "sig").write_bytes(der)
+    res = subprocess.run(["openssl", "dgst", "-sha256", "-verify", str(pub),
+                          "-signature", str(tmp_path / "sig"), str(tmp_path / "m")],
+                         capture_output=True)
+    assert res.returncode == 0, res.stdout + res.stderr
+
+    tok = chatbot_test_jwt(key.read_text(), audience="chatbot")
+    h, b, sg = tok.split(".")
+    hdr = json.loads(base64.urlsafe_b64decode(h + "=="))
+    assert hdr["alg"] == "ES256"
+    assert len(base64.urlsafe_b64decode(sg + "==")) == 64  # raw r||s
+
+
+def test_github_app_repo_listing_and_reaction(tmp_path):
+    from code_intelligence_amd.gh.github_app import GitHubApp
+    import subprocess as sp
+    key = tmp_path / "rsa.pem"
+    sp.run(["openssl", "genrsa", "-out", str(key), "2048"], check=True,
+           capture_output=True)
+    sess = FakeSession([
+        FakeResponse(200, {"token": "tkn", "expires_at": "2099-01-01T00:00:00Z"}),
+        FakeResponse(200, {"repositories": [{"full_name": "o/r"}]}),
+        FakeResponse(200, {"id": 1, "content": "+1"}),
+    ])
+    app = GitHubApp(pem_path=str(key), app_id="7", session=sess)
+    repos = app.list_installation_repos(42)
+    assert repos == [{"full_name": "o/r"}]
+    out = app.add_reaction("o", "r", 5, "+1", token="tkn")
+    assert out["content"] == "+1"
