@@ -144,3 +144,59 @@ def test_native_extension_is_loaded():
     lib = extension.require()
     assert lib.__file__.endswith(".so")
     assert "code_intelligence_amd" in lib.__file__
+
+
+@pytest.mark.parametrize("mode", ["lib", "fused"])
+def test_lstm_tiny_shapes(mode):
+    """Edge grids: B=1 (heavy M-tail), T=1, odd batch."""
+    from code_intelligence_amd.ops.lstm import lstm_forward, _cpu_lstm_loop
+    os.environ["CI_LSTM_MODE"] = mode
+    for B, T, In, H in [(1, 3, 16, 32), (3, 1, 32, 64), (5, 2, 24, 40)]:
+        torch.manual_seed(B * 100 + T)
+        x = torch.randn(B, T, In)
+        w_ih = torch.randn(4 * H, In) * 0.2
+        w_hh = torch.randn(4 * H, H) * 0.2
+        b_ih = torch.randn(4 * H) * 0.1
+        b_hh = torch.randn(4 * H) * 0.1
+        ref, h_ref, c_ref = _cpu_lstm_loop(
+            x, torch.zeros(B, H), torch.zeros(B, H), w_ih, w_hh, b_ih, b_hh)
+        dt = torch.bfloat16 if mode == "fused" else torch.float32
+        out, (hT, cT) = lstm_forward(
+            x.to(DEV, dt), torch.zeros(B, H, device=DEV, dtype=dt),
+            torch.zeros(B, H, device=DEV, dtype=dt),
+            w_ih.to(DEV, dt), w_hh.to(DEV, dt),
+            b_ih.to(DEV, dt), b_hh.to(DEV, dt))
+        tol = 0.06 if dt == torch.bfloat16 else 3e-4
+        assert torch.allclose(out.float().cpu(), ref, atol=tol), \
+            (B, T, (out.float().cpu() - ref).abs().max())
+
+
+def test_concat_pool_length_one_and_full():
+    from code_intelligence_amd.ops.pool import concat_pool, _cpu_concat_pool
+    torch.manual_seed(0)
+    h = torch.randn(2, 5, 8)
+    lengths = torch.tensor([1, 5])
+    ref = _cpu_concat_pool(h, lengths)
+    got = concat_pool(h.to(DEV), lengths.to(DEV)).cpu()
+    assert torch.allclose(got, ref, atol=1e-5)
+
+
+def test_ce_chunk_boundary():
+    """N not divisible by CHUNK and N < CHUNK both exercised."""
+    import code_intelligence_amd.ops.crossentropy as ce
+    from code_intelligence_amd.ops.crossentropy import tied_decoder_ce
+    import torch.nn.functional as F
+    old = ce._FusedCEFunction.CHUNK
+    try:
+        ce._FusedCEFunction.CHUNK = 7
+        os.environ["CI_CE_CHUNK"] = "7"
+        torch.manual_seed(0)
+        h = torch.randn(23, 16, device=DEV)
+        w = torch.randn(50, 16, device=DEV) * 0.3
+        t = torch.randint(0, 50, (23,), device=DEV)
+        loss = tied_decoder_ce(h, w, None, t)
+        ref = F.cross_entropy(F.linear(h, w), t)
+        assert abs(float(loss) - float(ref)) < 1e-4
+    finally:
+        ce._FusedCEFunction.CHUNK = old
+        os.environ.pop("CI_CE_CHUNK", None)
