@@ -143,8 +143,9 @@ class InferenceWrapper:
                     cur_bs //= 2
                     if cur_bs < 1:
                         raise
+            pooled_np = pooled.cpu().numpy()  # one D2H copy per batch
             for r, j in enumerate(idxs):
-                out[j] = pooled[r].cpu().numpy()
+                out[j] = pooled_np[r]
             i += len(idxs)
         return out
 
