@@ -1,0 +1,120 @@
+"""Coverage for CLIs, callbacks, schedules (table-driven, offline)."""
+import json
+import math
+
+import pytest
+import torch
+
+from code_intelligence_amd.control.automl_cli import main as automl_main
+from code_intelligence_amd.control.registry import LocalModelRegistry
+from code_intelligence_amd.label.cli import main as label_cli_main
+from code_intelligence_amd.train.callbacks import (CSVLogger, EarlyStopping,
+                                                   JSONRunLogger,
+                                                   ReduceLROnPlateau)
+from code_intelligence_amd.train.schedules import FlatSchedule, OneCycle
+from code_intelligence_amd.train.train_cli import main as train_main
+
+
+class _T:  # minimal trainer stand-in for callbacks
+    lr_scale = 1.0
+    model = None
+
+
+def test_one_cycle_shape():
+    s = OneCycle(max_lr=1.0, pct_start=0.3, div=25)
+    lr0, mom0 = s.at(0.0)
+    lr_peak, mom_peak = s.at(0.3)
+    lr_end, mom_end = s.at(1.0)
+    assert lr0 == pytest.approx(1.0 / 25)
+    assert lr_peak == pytest.approx(1.0)
+    assert lr_end < 1e-4                      # anneal to max_lr/final_div
+    assert mom0 == pytest.approx(0.95)
+    assert mom_peak == pytest.approx(0.85)
+    assert mom_end == pytest.approx(0.95)
+    # monotone up then down
+    ups = [s.at(f)[0] for f in (0.0, 0.1, 0.2, 0.3)]
+    downs = [s.at(f)[0] for f in (0.3, 0.5, 0.8, 1.0)]
+    assert ups == sorted(ups)
+    assert downs == sorted(downs, reverse=True)
+    assert FlatSchedule(0.5).at(0.7) == (0.5, None)
+
+
+def test_early_stopping_patience():
+    es = EarlyStopping(patience=2)
+    t = _T()
+    assert not es.on_epoch_end(t, 0, {"valid_loss": 1.0})
+    assert not es.on_epoch_end(t, 1, {"valid_loss": 1.1})
+    assert not es.on_epoch_end(t, 2, {"valid_loss": 1.2})
+    assert es.on_epoch_end(t, 3, {"valid_loss": 1.3})  # patience exceeded
+    # improvement resets
+    es2 = EarlyStopping(patience=1)
+    es2.on_epoch_end(t, 0, {"valid_loss": 1.0})
+    es2.on_epoch_end(t, 1, {"valid_loss": 1.5})
+    assert not es2.on_epoch_end(t, 2, {"valid_loss": 0.5})
+
+
+def test_reduce_lr_on_plateau():
+    cb = ReduceLROnPlateau(patience=1, factor=0.5)
+    t = _T()
+    cb.on_epoch_end(t, 0, {"valid_loss": 1.0})
+    cb.on_epoch_end(t, 1, {"valid_loss": 1.2})
+    assert t.lr_scale == 1.0
+    cb.on_epoch_end(t, 2, {"valid_loss": 1.3})
+    assert t.lr_scale == 0.5
+
+
+def test_csv_and_json_loggers(tmp_path):
+    csv_cb = CSVLogger(tmp_path / "h.csv")
+    run_cb = JSONRunLogger(tmp_path / "r.jsonl", config={"lr": 1}, every=2)
+    t = _T()
+    run_cb.on_train_begin(t)
+    run_cb.on_step_end(t, 2, 0.5)
+    run_cb.on_step_end(t, 3, 0.4)  # not logged (every=2)
+    csv_cb.on_epoch_end(t, 0, {"train_loss": 1.0})
+    run_cb.on_epoch_end(t, 0, {"train_loss": 1.0})
+    assert "train_loss" in (tmp_path / "h.csv").read_text()
+    lines = [json.loads(l) for l in open(tmp_path / "r.jsonl")]
+    assert lines[0]["event"] == "run_begin"
+    assert sum(1 for l in lines if l["event"] == "step") == 1
+
+
+def test_train_cli_end_to_end(tmp_path, capsys):
+    m = train_main(["--data_path", "synthetic:60", "--emb_sz", "16",
+                    "--n_hid", "24", "--n_layers", "2", "--vocab_sz", "200",
+                    "--bs", "4", "--bptt", "16", "--epochs", "1",
+                    "--model_path", str(tmp_path), "--dtype", "fp32"])
+    assert "valid_loss" in m
+    assert (tmp_path / "best_enc.pth").exists()   # SaveModel encoder artifact
+    assert (tmp_path / "history.csv").exists()
+    sd = torch.load(tmp_path / "best_enc.pth", weights_only=True)
+    assert "encoder.weight" in sd                 # fastai layout
+
+
+def test_automl_cli_roundtrip(tmp_path, capsys):
+    reg_dir = str(tmp_path / "reg")
+    reg = LocalModelRegistry(reg_dir)
+    rec = reg.create_training("d")
+    reg.finish_training(rec.name)
+    automl_main(["--registry", reg_dir, "get", "--dataset", "d"])
+    out = json.loads(capsys.readouterr().out)
+    assert out["latest_trained"]["name"] == rec.name
+    automl_main(["--registry", reg_dir, "deploy", "--name", rec.name])
+    assert json.loads(capsys.readouterr().out)["deployed"] is True
+    automl_main(["--registry", reg_dir, "is-training", "--dataset", "d"])
+    assert json.loads(capsys.readouterr().out) == {"isTraining": False}
+
+
+def test_label_cli_publish_and_logs(tmp_path, capsys):
+    spool = tmp_path / "spool.jsonl"
+    label_cli_main(["label-issue", "--issue", "kubeflow/kf#7",
+                    "--spool", str(spool)])
+    assert "published" in capsys.readouterr().out
+    from code_intelligence_amd.label.queueing import LocalQueue
+    msg = LocalQueue(spool_path=str(spool)).pull(0.2)
+    assert msg.attributes["repo_owner"] == "kubeflow"
+    assert msg.attributes["issue_num"] == "7"
+    logf = tmp_path / "w.jsonl"
+    logf.write_text(json.dumps({"time": "t", "level": "INFO",
+                                "message": "hello", "repo": "kf"}) + "\n")
+    label_cli_main(["logs", "--path", str(logf)])
+    assert "hello" in capsys.readouterr().out
