@@ -114,6 +114,23 @@ class LMTrainer:
         vl = tot / max(n, 1)
         return {"valid_loss": vl, "valid_ppl": math.exp(min(vl, 30.0))}
 
+    # --- checkpoint/resume (full training state; the reference only
+    # checkpoints model weights via fastai SaveModelCallback) -------------
+    def save_checkpoint(self, path) -> None:
+        torch.save({
+            "model": self.model.state_dict(),
+            "optimizer": self.opt.state_dict(),
+            "lr_scale": self.lr_scale,
+            "global_step": self.global_step,
+        }, path)
+
+    def load_checkpoint(self, path, map_location="cpu") -> None:
+        ckpt = torch.load(path, map_location=map_location, weights_only=False)
+        self.model.load_state_dict(ckpt["model"])
+        self.opt.load_state_dict(ckpt["optimizer"])
+        self.lr_scale = ckpt.get("lr_scale", 1.0)
+        self.global_step = ckpt.get("global_step", 0)
+
     def fit(self, train_loader, valid_loader=None, epochs: int = 1,
             one_cycle: Optional[bool] = None) -> dict:
         cfg = self.cfg
