@@ -70,3 +70,25 @@ def test_synthetic_tokens_in_range():
     docs = synthetic_issue_tokens(10, vocab_sz=1000, seed=0)
     flat = [t for d in docs for t in d]
     assert min(flat) >= 9 and max(flat) < 1000
+
+
+def test_prepare_data_script_end_to_end(tmp_path):
+    """scripts/prepare_data.py: archive -> docs.pt + vocab.json the train
+    CLI consumes (single-process path)."""
+    import subprocess
+    import sys
+    out = tmp_path / "corpus"
+    r = subprocess.run(
+        [sys.executable, "scripts/prepare_data.py", "--archive", "synthetic:40",
+         "--out", str(out), "--max_vocab", "500", "--workers", "1"],
+        capture_output=True, text=True, cwd=".")
+    assert r.returncode == 0, r.stderr
+    docs = torch.load(out / "docs.pt", weights_only=True)
+    assert len(docs) == 40
+    from code_intelligence_amd.text.tokenizer import Vocab
+    v = Vocab.load(out / "vocab.json")
+    assert all(0 <= t < len(v) for d in docs for t in d)
+    # consumable by the train CLI loader
+    from code_intelligence_amd.train.train_cli import load_docs
+    docs2, vsz = load_docs(str(out), 0)
+    assert vsz == len(v) and len(docs2) == 40
