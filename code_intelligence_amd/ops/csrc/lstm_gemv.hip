@@ -1,0 +1,100 @@
+// K2 serve variant: fused GEMV + cell kernel for small batches
+// (SURVEY.md §2.4 "persistent-LSTM variant for small B serve path").
+//
+// At B<=8 (single-request serving) the recurrent step is a GEMV bound by
+// streaming the 46 MB weight matrix; tiling for MFMA is pointless. One
+// block per hidden unit j: its four waves compute the four gate dot
+// products <h_b, W[g*H+j,:]> with lane-split K (coalesced 16-B row reads,
+// wave shfl reduction), then lane 0 finishes the cell and stores
+// h/c/gates — a single launch per timestep replaces hipBLASLt GEMV +
+// pointwise kernel. Grid = H blocks (2400 at the deployed shape) fills
+// all 256 CUs; batch rows loop inside the block (B tiny).
+#include "common.h"
+
+namespace ci {
+
+typedef __bf16 bf16x8g __attribute__((ext_vector_type(8)));
+
+__global__ __launch_bounds__(256) void lstm_cell_gemv(
+    const __hip_bfloat16* __restrict__ h_prev, long h_rs,
+    const __hip_bfloat16* __restrict__ w_hh,   // (4H, H) row-major
+    const __hip_bfloat16* __restrict__ xp, long xp_rs,
+    const float* __restrict__ bias,
+    const float* __restrict__ c_prev, long cp_rs,
+    __hip_bfloat16* __restrict__ h_out, long ho_rs,
+    float* __restrict__ c_out, long co_rs,
+    __hip_bfloat16* __restrict__ gates_out, long go_rs,
+    int B, int H) {
+  const int j = blockIdx.x;          // hidden unit
+  const int wave = threadIdx.x >> 6; // gate g in {i,f,g,o}
+  const int lane = threadIdx.x & 63;
+  const __hip_bfloat16* wrow = w_hh + (long)(wave * H + j) * H;
+  __shared__ float dots[4];
+  const int Hv = H / 8 * 8;
+  for (int b = 0; b < B; ++b) {
+    const __hip_bfloat16* hrow = h_prev + (long)b * h_rs;
+    float acc = 0.f;
+    for (int k = lane * 8; k < Hv; k += 64 * 8) {
+      bf16x8g wv = *reinterpret_cast<const bf16x8g*>(wrow + k);
+      bf16x8g hv = *reinterpret_cast<const bf16x8g*>(hrow + k);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e)
+        acc += (float)wv[e] * (float)hv[e];
+    }
+    for (int k = Hv + lane; k < H; k += 64)
+      acc += ld(wrow + k) * ld(hrow + k);
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      acc += __shfl_down(acc, off);
+    if (lane == 0) dots[wave] = acc;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const long xo = (long)b * xp_rs + j;
+      const float gi = sigmoidf_(dots[0] + ld(xp + xo) + bias[j]);
+      const float gf = sigmoidf_(dots[1] + ld(xp + xo + H) + bias[j + H]);
+      const float gg = tanhf(dots[2] + ld(xp + xo + 2 * H) + bias[j + 2 * H]);
+      const float go = sigmoidf_(dots[3] + ld(xp + xo + 3 * H) + bias[j + 3 * H]);
+      const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
+      const float h = go * tanhf(c);
+      st(h_out + (long)b * ho_rs + j, h);
+      c_out[(long)b * co_rs + j] = c;
+      const long g0 = (long)b * go_rs + j;
+      st(gates_out + g0, gi);
+      st(gates_out + g0 + H, gf);
+      st(gates_out + g0 + 2 * H, gg);
+      st(gates_out + g0 + 3 * H, go);
+    }
+    __syncthreads();
+  }
+}
+
+// whole-sequence driver over time-major (T,B,·) saves, one launch/step.
+void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                           at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
+                           at::Tensor cs, at::Tensor gates) {
+  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
+  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates);
+  TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16,
+              "gemv cell kernel is bf16");
+  const int T = xp.size(0), B = xp.size(1);
+  const int H = w_hh.size(1);
+  auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
+  auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
+  auto* gp = reinterpret_cast<__hip_bfloat16*>(gates.data_ptr());
+  auto* wp = reinterpret_cast<const __hip_bfloat16*>(w_hh.data_ptr());
+  auto h0c = h0.contiguous();
+  auto* h0p = reinterpret_cast<const __hip_bfloat16*>(h0c.data_ptr());
+  for (int t = 0; t < T; ++t) {
+    const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * B * H;
+    const float* cp = (t == 0) ? c0.data_ptr<float>()
+                               : cs.data_ptr<float>() + (long)(t - 1) * B * H;
+    hipLaunchKernelGGL(lstm_cell_gemv, dim3(H), dim3(256), 0, stream(),
+        hp, (long)H, wp, xpp + (long)t * B * 4 * H, (long)4 * H,
+        bias.data_ptr<float>(), cp, (long)H,
+        hsp + (long)t * B * H, (long)H,
+        cs.data_ptr<float>() + (long)t * B * H, (long)H,
+        gp + (long)t * B * 4 * H, (long)4 * H, B, H);
+  }
+}
+
+}  // namespace ci

@@ -30,6 +30,7 @@ SOURCES = [
     "bindings.cpp",
     "lstm_pointwise.hip",
     "lstm_gemm.hip",
+    "lstm_gemv.hip",
     "pool.hip",
     "adam.hip",
     "ce.hip",

@@ -78,7 +78,12 @@ class _FusedLSTMFunction(torch.autograd.Function):
         # (471 vs 492 ms/step at the deployed shape, profiles/BENCH_HISTORY.md);
         # CI_LSTM_MODE=fused selects the hand-written fused path.
         mode = os.environ.get("CI_LSTM_MODE", "lib")
-        if mode == "fused" and dt == torch.bfloat16 and H % 8 == 0:
+        if dt == torch.bfloat16 and B <= 8 and mode != "lib-only":
+            # serve/decode regime: weight-streaming-bound GEMV — the fused
+            # GEMV+cell kernel replaces hipBLASLt GEMV + pointwise launch
+            lib.lstm_seq_forward_gemv(xp, bias, h0, c0.to(torch.float32),
+                                      w_hh, hs, cs, gates)
+        elif mode == "fused" and dt == torch.bfloat16 and H % 8 == 0:
             lib.lstm_seq_forward_fused(xp, bias, h0, c0.to(torch.float32), w_hh,
                                        hs, cs, gates)
         else:
