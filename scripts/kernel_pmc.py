@@ -60,6 +60,15 @@ scale = torch.full((1,), 1e-5, device=dev)
 for _ in range(3):
     lib.ce_dlogits(logits, tgt, b32, lse, scale)
 
+# serve GEMV+cell kernel (B=1)
+xp1 = torch.randn(T, 1, 4 * H, device=dev, dtype=dt)
+h1 = torch.randn(1, H, device=dev, dtype=dt) * 0.1
+c1 = torch.randn(1, H, device=dev, dtype=torch.float32) * 0.1
+hs1 = torch.empty(T, 1, H, device=dev, dtype=dt)
+cs1 = torch.empty(T, 1, H, device=dev, dtype=torch.float32)
+g1 = torch.empty(T, 1, 4 * H, device=dev, dtype=dt)
+lib.lstm_seq_forward_gemv(xp1, bias, h1, c1, w, hs1, cs1, g1)
+
 # concat-pool (K5) at serve shape
 hid = torch.randn(200, 512, 800, device=dev, dtype=dt)
 lens = torch.randint(1, 513, (200,), device=dev, dtype=torch.int32)
