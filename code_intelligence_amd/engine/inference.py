@@ -13,7 +13,6 @@ per-timestep launch overhead that dominates at serve batch sizes.
 from __future__ import annotations
 
 import json
-import math
 from pathlib import Path
 from typing import List, Optional, Sequence
 

@@ -47,8 +47,6 @@ class RepoSpecificLabelModel(IssueLabelModel):
         """reference repo_specific_model.py:32-88."""
         store = store or default_store()
         cfg = RepoConfig(repo_owner, repo_name)
-        import io
-        import pickle
         import tempfile
         with tempfile.NamedTemporaryFile(suffix=".dpkl") as tmp:
             store.download(cfg.model_gcs_uri, tmp.name)

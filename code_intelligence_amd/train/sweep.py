@@ -28,7 +28,7 @@ import subprocess
 import sys
 import time
 from pathlib import Path
-from typing import Dict, Iterator, List
+from typing import Dict, List
 
 import yaml
 
