@@ -36,7 +36,41 @@ from torch import Tensor
 
 from . import extension as ext
 
-__all__ = ["lstm_forward"]
+__all__ = ["lstm_forward", "sync_dw_stream"]
+
+_side_stream = None
+
+
+def _side() -> "torch.cuda.Stream":
+    global _side_stream
+    if _side_stream is None:
+        _side_stream = torch.cuda.Stream()
+    return _side_stream
+
+
+def _side_dw_enabled() -> bool:
+    """Opt-in (CI_SIDE_DW=1): compute dW_ih/db on a second HIP stream so
+    they overlap the NEXT layer's sequential backward loop (which leaves
+    ~40% of the chip idle). Only parameters with no other graph consumers
+    (w_ih, b_ih, b_hh) are accumulated manually; dW_hh must return through
+    autograd (the weight-drop mask backward consumes it). Anything that
+    reads .grad after backward must call sync_dw_stream() first — the
+    trainer does."""
+    return os.environ.get("CI_SIDE_DW", "0") == "1"
+
+
+def sync_dw_stream() -> None:
+    """Order the side-stream weight-grad GEMMs before the current stream.
+    No-op unless CI_SIDE_DW work was launched."""
+    if _side_stream is not None and torch.cuda.is_available():
+        torch.cuda.current_stream().wait_stream(_side_stream)
+
+
+def _accum_grad(p: Tensor, g: Tensor) -> None:
+    if p.grad is None:
+        p.grad = g.detach().to(p.dtype)
+    else:
+        p.grad.add_(g.to(p.dtype))
 
 
 def _cpu_lstm_loop(x: Tensor, h0: Tensor, c0: Tensor, w_ih: Tensor, w_hh: Tensor,
@@ -90,6 +124,7 @@ class _FusedLSTMFunction(torch.autograd.Function):
             lib.lstm_seq_forward_lib(xp, bias, h0, c0.to(torch.float32), w_hh,
                                      hs, cs, gates)
         ctx.save_for_backward(x_tm, h0, c0, w_ih, w_hh, gates, hs, cs)
+        ctx.direct_params = (w_ih, b_ih, b_hh)  # no other graph consumers
         hT = hs[-1].clone()
         cT = cs[-1].to(dt)
         out = hs.transpose(0, 1)  # (B,T,H) view of time-major storage
@@ -112,13 +147,28 @@ class _FusedLSTMFunction(torch.autograd.Function):
                               dgates, dh0, dc0)
         dg2 = dgates.view(T * B, 4 * H)
         dx_tm = torch.mm(dg2, w_ih).view(T, B, In)
-        dw_ih = torch.mm(dg2.t(), x_tm.view(T * B, In))
         # dW_hh = sum_t h_{t-1}^T dgates_t — time-major slices stay
         # contiguous, so no (T,B,H) cat materializes:
         dw_hh = torch.mm(dgates[0].t(), h0.to(dt))
         if T > 1:
             dw_hh += torch.mm(dgates[1:].reshape((T - 1) * B, 4 * H).t(),
                               hs[:-1].reshape((T - 1) * B, H))
+        if _side_dw_enabled():
+            # overlap dW_ih/db with the next layer's sequential loop
+            pw, pbi, pbh = ctx.direct_params
+            s_ = _side()
+            s_.wait_stream(torch.cuda.current_stream())
+            dg2.record_stream(s_)
+            x_tm.record_stream(s_)
+            with torch.cuda.stream(s_):
+                dw_ih_s = torch.mm(dg2.t(), x_tm.view(T * B, In))
+                db_s = dg2.sum(dim=0)
+                _accum_grad(pw, dw_ih_s)
+                _accum_grad(pbi, db_s)
+                _accum_grad(pbh, db_s)
+            return (dx_tm.transpose(0, 1), dh0.to(dt), dc0.to(dt),
+                    None, dw_hh.to(w_hh.dtype), None, None)
+        dw_ih = torch.mm(dg2.t(), x_tm.view(T * B, In))
         db = dg2.sum(dim=0).to(dt)
         return (dx_tm.transpose(0, 1), dh0.to(dt), dc0.to(dt),
                 dw_ih.to(w_ih.dtype), dw_hh.to(w_hh.dtype), db, db.clone())

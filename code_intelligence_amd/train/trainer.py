@@ -92,6 +92,9 @@ class LMTrainer:
         self.opt.zero_grad(set_to_none=True)
         loss = self.loss_on_batch(x, y)
         loss.backward()
+        if x.is_cuda:
+            from ..ops.lstm import sync_dw_stream
+            sync_dw_stream()  # side-stream dW grads (CI_SIDE_DW) ordered
         if self.dist is not None:
             self.dist.finalize()
         if self.cfg.clip:

@@ -143,3 +143,39 @@ def test_graphed_encoder_matches_eager():
     ids2 = torch.randint(9, 400, (B, T), device=DEV)
     got2 = g.run(ids2).float()
     assert not torch.allclose(got2, eager, atol=1e-3)
+
+
+def test_side_dw_grads_match():
+    """CI_SIDE_DW=1 (side-stream dW_ih/db) produces identical grads."""
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.ops.lstm import sync_dw_stream
+    from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
+
+    def run(side):
+        if side:
+            os.environ["CI_SIDE_DW"] = "1"
+        else:
+            os.environ.pop("CI_SIDE_DW", None)
+        try:
+            torch.manual_seed(3)
+            m = AWDLSTM(vocab_sz=300, emb_sz=32, n_hid=48, n_layers=3,
+                        output_p=0, hidden_p=0, input_p=0, embed_p=0,
+                        weight_p=0).to(DEV)
+            tr = LMTrainer(m, TrainConfig(alpha=0, beta=0))
+            m.eval()
+            g = torch.Generator().manual_seed(5)
+            x = torch.randint(9, 300, (4, 12), generator=g).to(DEV)
+            y = torch.roll(x, -1, 1)
+            tr.loss_on_batch(x, y).backward()
+            sync_dw_stream()
+            return {n: p.grad.float().cpu() for n, p in m.named_parameters()
+                    if p.grad is not None}
+        finally:
+            os.environ.pop("CI_SIDE_DW", None)
+
+    base = run(False)
+    side = run(True)
+    assert set(base) == set(side)
+    for n in base:
+        assert torch.allclose(base[n], side[n], atol=1e-5), \
+            (n, (base[n] - side[n]).abs().max())
