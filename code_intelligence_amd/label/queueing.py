@@ -99,6 +99,41 @@ class LocalQueue(BaseQueue):
             return None
 
 
+class LocalBroker:
+    """Idempotent topic/subscription management (reference: pubsub_util.py
+    check_subscription_name_exists / create_subscription_if_not_exists /
+    create_topic_if_not_exists) over a local spool directory: one queue
+    (spool file) per subscription."""
+
+    def __init__(self, root):
+        self.root = Path(root)
+        self.root.mkdir(parents=True, exist_ok=True)
+
+    def _topic_dir(self, topic: str) -> Path:
+        return self.root / topic
+
+    def check_topic_exists(self, topic: str) -> bool:
+        return self._topic_dir(topic).is_dir()
+
+    def create_topic_if_not_exists(self, topic: str) -> None:
+        self._topic_dir(topic).mkdir(parents=True, exist_ok=True)
+
+    def check_subscription_name_exists(self, topic: str, sub: str) -> bool:
+        return (self._topic_dir(topic) / f"{sub}.jsonl").exists()
+
+    def create_subscription_if_not_exists(self, topic: str, sub: str
+                                          ) -> "LocalQueue":
+        self.create_topic_if_not_exists(topic)
+        spool = self._topic_dir(topic) / f"{sub}.jsonl"
+        spool.touch(exist_ok=True)
+        return LocalQueue(spool_path=str(spool))
+
+    def publish(self, topic: str, data: bytes = b"", **attributes) -> None:
+        """Fan out to every subscription of the topic."""
+        for spool in self._topic_dir(topic).glob("*.jsonl"):
+            LocalQueue(spool_path=str(spool)).publish(data, **attributes)
+
+
 def queue_from_env() -> BaseQueue:
     """Reference env contract (worker.py:68-86): PROJECT + ISSUE_EVENT_TOPIC
     + ISSUE_EVENT_SUBSCRIPTION select Pub/Sub; otherwise a LocalQueue

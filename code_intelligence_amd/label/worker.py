@@ -62,6 +62,33 @@ class GitHubIssueClient:
         return r.json()
 
 
+def wait_for_endpoint(url: str, session=None, timeout_s: float = 120.0,
+                      base_delay_s: float = 1.0, max_delay_s: float = 30.0
+                      ) -> bool:
+    """Exponential-backoff wait for a dependency's /healthz (the analogue
+    of the reference's wait_for_gcp_account retry loop, worker.py:446-463:
+    dependencies come up in any order; the worker retries with backoff
+    instead of crash-looping)."""
+    import time
+    if session is None:
+        import requests
+        session = requests.Session()
+    deadline = time.monotonic() + timeout_s
+    delay = base_delay_s
+    while True:
+        try:
+            r = session.get(url.rstrip("/") + "/healthz", timeout=5)
+            if r.status_code == 200:
+                return True
+        except Exception:
+            pass
+        if time.monotonic() >= deadline:
+            return False
+        log.info("dependency %s not ready; retrying in %.1fs", url, delay)
+        time.sleep(delay)
+        delay = min(delay * 2, max_delay_s)
+
+
 class Worker:
     BOT_MARKER = "<!-- issue-label-bot -->"
 

@@ -270,3 +270,45 @@ def test_local_queue_spool_roundtrip(tmp_path):
     msg = cons.pull(timeout=0.2)
     assert msg is not None
     assert msg.attributes["issue_num"] == "3"
+
+
+def test_local_broker_idempotent(tmp_path):
+    from code_intelligence_amd.label.queueing import LocalBroker
+    b = LocalBroker(tmp_path)
+    assert not b.check_topic_exists("events")
+    b.create_topic_if_not_exists("events")
+    b.create_topic_if_not_exists("events")  # idempotent
+    assert b.check_topic_exists("events")
+    assert not b.check_subscription_name_exists("events", "worker")
+    q1 = b.create_subscription_if_not_exists("events", "worker")
+    q2 = b.create_subscription_if_not_exists("events", "worker2")
+    assert b.check_subscription_name_exists("events", "worker")
+    b.publish("events", repo_owner="o", repo_name="r", issue_num=1)
+    m1, m2 = q1.pull(0.2), q2.pull(0.2)  # fan-out to both subscriptions
+    assert m1.attributes["issue_num"] == "1"
+    assert m2.attributes["issue_num"] == "1"
+
+
+def test_wait_for_endpoint_backoff():
+    from code_intelligence_amd.label.worker import wait_for_endpoint
+
+    class FlakySession:
+        def __init__(self):
+            self.calls = 0
+
+        def get(self, url, timeout=None):
+            self.calls += 1
+            class R:
+                status_code = 503 if self.calls < 3 else 200
+            return R()
+
+    s = FlakySession()
+    assert wait_for_endpoint("http://svc", session=s, timeout_s=30,
+                             base_delay_s=0.01)
+    assert s.calls == 3
+
+    class DownSession:
+        def get(self, url, timeout=None):
+            raise ConnectionError()
+    assert not wait_for_endpoint("http://svc", session=DownSession(),
+                                 timeout_s=0.05, base_delay_s=0.01)
