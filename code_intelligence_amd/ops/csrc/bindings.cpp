@@ -1,5 +1,6 @@
 // Python bindings for the code_intelligence_amd gfx950 kernels.
 #include <torch/extension.h>
+#include <string>
 #include <vector>
 
 namespace ci {
@@ -26,6 +27,9 @@ void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                  std::vector<at::Tensor> eass, double lr, double b1, double b2,
                  double eps, double wd, double bc1, double bc2);
 at::Tensor emb_gather(at::Tensor weight, at::Tensor ids, at::Tensor rowmask);
+std::vector<std::string> tokenize_core(const std::string& text);
+std::vector<std::vector<std::string>> tokenize_core_batch(
+    const std::vector<std::string>& texts);
 at::Tensor emb_scatter(at::Tensor gout, at::Tensor ids, at::Tensor rowmask,
                        long V, long pad_idx);
 }  // namespace ci
@@ -44,5 +48,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_dlogits", &ci::ce_dlogits, "in-place (softmax-onehot)*scale");
   m.def("fused_adamw", &ci::fused_adamw, "fused AdamW step");
   m.def("emb_gather", &ci::emb_gather, "embedding gather with row dropout");
+  m.def("tokenize_core", &ci::tokenize_core, "native ASCII tokenizer core");
+  m.def("tokenize_core_batch", &ci::tokenize_core_batch,
+        "native ASCII tokenizer core (batch, GIL released)");
   m.def("emb_scatter", &ci::emb_scatter, "embedding grad scatter-add");
 }

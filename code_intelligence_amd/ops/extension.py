@@ -35,6 +35,7 @@ SOURCES = [
     "adam.hip",
     "ce.hip",
     "embedding.hip",
+    "tokenizer.cpp",
 ]
 
 

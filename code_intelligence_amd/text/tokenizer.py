@@ -129,14 +129,40 @@ default_post_rules = [replace_all_caps, deal_caps]
 
 
 class Tokenizer:
-    """pre_rules (str->str) -> regex word split -> post_rules (tokens->tokens)."""
+    """pre_rules (str->str) -> regex word split -> post_rules (tokens->tokens).
 
-    def __init__(self, pre_rules=None, post_rules=None, markdown: bool = True):
+    With ``native=True`` (default) and default rules, pure-ASCII documents
+    run through the C++ tokenizer core (ops/csrc/tokenizer.cpp — token-exact
+    for replace_rep/replace_wrep/split/caps; fix_html + markdown
+    normalization stay in Python); non-ASCII or custom-rule inputs use the
+    Python path, so results are identical either way."""
+
+    def __init__(self, pre_rules=None, post_rules=None, markdown: bool = True,
+                 native: bool = True):
         self.pre_rules = pre_rules if pre_rules is not None else (
             markdown_pre_rules if markdown else default_pre_rules)
         self.post_rules = post_rules if post_rules is not None else default_post_rules
+        self._default_rules = pre_rules is None and post_rules is None
+        self._native = None
+        if native and self._default_rules:
+            try:
+                from ..ops import extension as _ext
+                lib = _ext.load(required=False)
+                if lib is not None and hasattr(lib, "tokenize_core"):
+                    self._native = lib
+            except Exception:
+                self._native = None
+        # string-level rules that precede the native core (everything except
+        # the token-level-equivalent rep/wrep/space rules it implements)
+        self._native_pre = ([markdown_rules] if markdown else []) + [fix_html]
 
     def process_text(self, text: str) -> List[str]:
+        if self._native is not None:
+            pre = text
+            for rule in self._native_pre:
+                pre = rule(pre)
+            if pre.isascii():
+                return self._native.tokenize_core(pre)
         for rule in self.pre_rules:
             text = rule(text)
         toks = _re_word.findall(text)
@@ -145,6 +171,22 @@ class Tokenizer:
         return toks
 
     def process_all(self, texts: Iterable[str]) -> List[List[str]]:
+        if self._native is not None:
+            pres, idx_native, out = [], [], {}
+            texts = list(texts)
+            for i, t in enumerate(texts):
+                pre = t
+                for rule in self._native_pre:
+                    pre = rule(pre)
+                if pre.isascii():
+                    pres.append(pre)
+                    idx_native.append(i)
+            if pres:
+                for i, toks in zip(idx_native,
+                                   self._native.tokenize_core_batch(pres)):
+                    out[i] = toks
+            return [out[i] if i in out else self.process_text(texts[i])
+                    for i in range(len(texts))]
         return [self.process_text(t) for t in texts]
 
 

@@ -117,3 +117,29 @@ def test_loader_reshuffles_between_epochs():
     assert not torch.equal(e1, e2)  # epoch-order reshuffle
     assert e1.numel() == e2.numel()  # same window count (tail drop is
     # order-dependent, like fastai's LMDataLoader)
+
+
+def test_native_tokenizer_parity():
+    """C++ tokenizer core is token-exact vs the Python rules on ASCII
+    (non-ASCII inputs transparently fall back to Python)."""
+    import random
+    import string
+    from code_intelligence_amd.text.tokenizer import Tokenizer
+    nat = Tokenizer(native=True)
+    py = Tokenizer(native=False)
+    if nat._native is None:
+        import pytest
+        pytest.skip("native extension not built")
+    rng = random.Random(1)
+    pool = string.ascii_letters + string.digits + " \n\t.,!?#/\\()[]{}`*->:;\"'"
+    for i in range(300):
+        s = "".join(rng.choice(pool) for _ in range(rng.randint(0, 250)))
+        if i % 4 == 0:
+            s += " soooooo COOL Cool go go go go go !!!!!! "
+        assert nat.process_text(s) == py.process_text(s), s[:120]
+    # non-ASCII routes through Python: identical by construction
+    s = "héllo wörld CRASH"
+    assert nat.process_text(s) == py.process_text(s)
+    # batch API equals per-doc API
+    docs = ["a b C", "x " * 30, "sooo COOL"]
+    assert nat.process_all(docs) == [nat.process_text(d) for d in docs]
