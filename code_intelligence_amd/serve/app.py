@@ -28,12 +28,29 @@ from ..engine.inference import InferenceWrapper
 log = logging.getLogger(__name__)
 
 
+def _resolve_model_path(model_path: str | None) -> str:
+    """Reference app.py:20-34 downloads the model artifact from GCS at pod
+    start; here gs:// URIs resolve through the object store into a local
+    cache dir."""
+    path = model_path or os.environ.get("MODEL_PATH", "model_files")
+    if not path.startswith("gs://"):
+        return path
+    import tempfile
+    from ..gh.gcs_util import default_store
+    store = default_store()
+    cache = os.path.join(tempfile.gettempdir(), "ci_model_cache")
+    os.makedirs(cache, exist_ok=True)
+    for name in ("config.json", "vocab.json", "encoder.pth"):
+        store.download(f"{path.rstrip('/')}/{name}", os.path.join(cache, name))
+    return cache
+
+
 def create_app(wrapper: InferenceWrapper | None = None,
                model_path: str | None = None) -> Flask:
     app = Flask("issue_embedding_server")
     if wrapper is None:
         wrapper = InferenceWrapper(
-            model_path=model_path or os.environ.get("MODEL_PATH", "model_files"),
+            model_path=_resolve_model_path(model_path),
             use_graphs=os.environ.get("CI_SERVE_GRAPHS", "0") == "1")
     app.config["wrapper"] = wrapper
 

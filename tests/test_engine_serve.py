@@ -69,3 +69,42 @@ def test_flask_text_contract(tmp_path):
     r2 = client.post("/texts", json={"documents": [
         {"title": "bug", "body": "w1"}, {"title": "feat", "body": "w2"}]})
     assert r2.get_json()["shape"] == [2, 48]
+
+
+def test_oom_backoff_halves_batch(tmp_path, monkeypatch):
+    """reference inference.py:214-223: CUDA OOM -> halve bs until it fits."""
+    w = _tiny_wrapper(tmp_path)
+    calls = []
+    real = w._encode_batch
+
+    def flaky(ids, lengths):
+        calls.append(ids.shape[0])
+        if ids.shape[0] > 1 and len(calls) < 3:
+            raise torch.cuda.OutOfMemoryError("fake OOM")
+        return real(ids, lengths)
+
+    monkeypatch.setattr(w, "_encode_batch", flaky)
+    monkeypatch.setattr(torch.cuda, "empty_cache", lambda: None)
+    out = w.texts_to_embedding(["w1 w2", "w3", "w4 w5 w6", "w7"], bs=4)
+    assert out.shape == (4, 48)
+    assert calls[0] == 4 and calls[1] == 2  # halved after OOM
+
+
+def test_serve_model_uri_resolution(tmp_path, monkeypatch):
+    """gs:// MODEL_PATH downloads artifacts through the object store."""
+    from code_intelligence_amd.gh.gcs_util import ObjectStore
+    import code_intelligence_amd.gh.gcs_util as gcs
+    from code_intelligence_amd.serve.app import _resolve_model_path
+    from code_intelligence_amd.engine.inference import save_artifacts
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+    store = ObjectStore(root=tmp_path / "store")
+    monkeypatch.setattr(gcs, "_default_store", store)
+    model = AWDLSTM(vocab_sz=20, emb_sz=8, n_hid=12, n_layers=1)
+    local = tmp_path / "art"
+    save_artifacts(model, Vocab(defaults_specials), local)
+    for name in ("config.json", "vocab.json", "encoder.pth"):
+        store.upload(str(local / name), f"gs://models/lm/{name}")
+    path = _resolve_model_path("gs://models/lm")
+    import json as _json
+    assert _json.loads(open(f"{path}/config.json").read())["emb_sz"] == 8
