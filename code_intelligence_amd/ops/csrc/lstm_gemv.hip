@@ -25,46 +25,66 @@ __global__ __launch_bounds__(256) void lstm_cell_gemv(
     float* __restrict__ c_out, long co_rs,
     __hip_bfloat16* __restrict__ gates_out, long go_rs,
     int B, int H) {
+  constexpr int BMAX = 8;            // dispatch guarantees B <= 8
   const int j = blockIdx.x;          // hidden unit
   const int wave = threadIdx.x >> 6; // gate g in {i,f,g,o}
   const int lane = threadIdx.x & 63;
   const __hip_bfloat16* wrow = w_hh + (long)(wave * H + j) * H;
-  __shared__ float dots[4];
+  __shared__ float dots[BMAX][4];
   const int Hv = H / 8 * 8;
-  for (int b = 0; b < B; ++b) {
-    const __hip_bfloat16* hrow = h_prev + (long)b * h_rs;
-    float acc = 0.f;
-    for (int k = lane * 8; k < Hv; k += 64 * 8) {
-      bf16x8g wv = *reinterpret_cast<const bf16x8g*>(wrow + k);
-      bf16x8g hv = *reinterpret_cast<const bf16x8g*>(hrow + k);
+  // read each W element ONCE; accumulate all batch rows simultaneously
+  // (h rows are tiny and L2-resident; W is the 46 MB stream)
+  float acc[BMAX];
+  #pragma unroll
+  for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
+  // acc[] must be register-resident: unroll over BMAX with a guard so
+  // every index is compile-time (runtime-indexed arrays go to scratch —
+  // CDNA guide §5.4 rule 20)
+  for (int k = lane * 8; k < Hv; k += 64 * 8) {
+    bf16x8g wv = *reinterpret_cast<const bf16x8g*>(wrow + k);
+    #pragma unroll
+    for (int b = 0; b < BMAX; ++b) {
+      if (b >= B) break;
+      bf16x8g hv = *reinterpret_cast<const bf16x8g*>(
+          h_prev + (long)b * h_rs + k);
       #pragma unroll
       for (int e = 0; e < 8; ++e)
-        acc += (float)wv[e] * (float)hv[e];
+        acc[b] += (float)wv[e] * (float)hv[e];
     }
-    for (int k = Hv + lane; k < H; k += 64)
-      acc += ld(wrow + k) * ld(hrow + k);
+  }
+  for (int k = Hv + lane; k < H; k += 64) {
+    const float wv = ld(wrow + k);
+    #pragma unroll
+    for (int b = 0; b < BMAX; ++b) {
+      if (b >= B) break;
+      acc[b] += wv * ld(h_prev + (long)b * h_rs + k);
+    }
+  }
+  #pragma unroll
+  for (int b = 0; b < BMAX; ++b) {
+    if (b >= B) break;
+    float a = acc[b];
     #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
-      acc += __shfl_down(acc, off);
-    if (lane == 0) dots[wave] = acc;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      const long xo = (long)b * xp_rs + j;
-      const float gi = sigmoidf_(dots[0] + ld(xp + xo) + bias[j]);
-      const float gf = sigmoidf_(dots[1] + ld(xp + xo + H) + bias[j + H]);
-      const float gg = tanhf(dots[2] + ld(xp + xo + 2 * H) + bias[j + 2 * H]);
-      const float go = sigmoidf_(dots[3] + ld(xp + xo + 3 * H) + bias[j + 3 * H]);
-      const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
-      const float h = go * tanhf(c);
-      st(h_out + (long)b * ho_rs + j, h);
-      c_out[(long)b * co_rs + j] = c;
-      const long g0 = (long)b * go_rs + j;
-      st(gates_out + g0, gi);
-      st(gates_out + g0 + H, gf);
-      st(gates_out + g0 + 2 * H, gg);
-      st(gates_out + g0 + 3 * H, go);
-    }
-    __syncthreads();
+      a += __shfl_down(a, off);
+    if (lane == 0) dots[b][wave] = a;
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < B; b += blockDim.x) {
+    const long xo = (long)b * xp_rs + j;
+    const float gi = sigmoidf_(dots[b][0] + ld(xp + xo) + bias[j]);
+    const float gf = sigmoidf_(dots[b][1] + ld(xp + xo + H) + bias[j + H]);
+    const float gg = tanhf(dots[b][2] + ld(xp + xo + 2 * H) + bias[j + 2 * H]);
+    const float go = sigmoidf_(dots[b][3] + ld(xp + xo + 3 * H) + bias[j + 3 * H]);
+    const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
+    const float h = go * tanhf(c);
+    st(h_out + (long)b * ho_rs + j, h);
+    c_out[(long)b * co_rs + j] = c;
+    const long g0 = (long)b * go_rs + j;
+    st(gates_out + g0, gi);
+    st(gates_out + g0 + H, gf);
+    st(gates_out + g0 + 2 * H, gg);
+    st(gates_out + g0 + 3 * H, go);
   }
 }
 
