@@ -312,3 +312,35 @@ def test_wait_for_endpoint_backoff():
             raise ConnectionError()
     assert not wait_for_endpoint("http://svc", session=DownSession(),
                                  timeout_s=0.05, base_delay_s=0.01)
+
+
+def test_feedback_collector(tmp_path):
+    from code_intelligence_amd.label.feedback import FeedbackCollector
+    bot_body = (Worker.BOT_MARKER + "\n"
+                "Issue-Label Bot is automatically applying the labels below:\n\n"
+                "| Label | Probability |\n|---|---|\n"
+                "| kind/bug | 0.91 |\n| area/ops | 0.66 |\n\nthanks!")
+
+    class Sess:
+        def get(self, url, params=None, headers=None):
+            class R:
+                status_code = 200
+
+                def raise_for_status(self):
+                    pass
+
+                def json(self):
+                    if url.endswith("/issues/7/comments"):
+                        return [{"body": "unrelated"},
+                                {"body": bot_body,
+                                 "reactions": {"+1": 3, "-1": 1}}]
+                    return []
+            return R()
+
+    fc = FeedbackCollector(session=Sess())
+    out = tmp_path / "fb.jsonl"
+    records = fc.collect("o", "r", [7, 8], output=str(out))
+    assert len(records) == 1
+    assert records[0]["labels"] == ["kind/bug", "area/ops"]
+    assert records[0]["score"] == 2
+    assert out.read_text().count("\n") == 1
