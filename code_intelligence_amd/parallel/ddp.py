@@ -95,6 +95,7 @@ class DistributedGrads:
         params = uniq
         self.buckets: List[_Bucket] = []
         self._param_bucket = {}
+        self._armed = False  # hooks no-op unless prepare() armed this pass
         if not self.enabled:
             return
         cap = int(bucket_mb * 1024 * 1024)
@@ -118,6 +119,8 @@ class DistributedGrads:
         self.buckets.append(_Bucket(params, p0.device, p0.dtype))
 
     def _hook(self, p: Tensor):
+        if not self._armed:
+            return  # gradient-accumulation micro-step: no communication
         b, off = self._param_bucket[id(p)]
         b.buffer[off: off + p.numel()].copy_(p.grad.reshape(-1))
         b.pending -= 1
@@ -125,13 +128,18 @@ class DistributedGrads:
             b.work = dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
                                      group=self.group, async_op=True)
 
-    def prepare(self):
+    def prepare(self, sync: bool = True):
+        """Arm the hooks for the coming backward. sync=False = this is a
+        gradient-accumulation micro-step (DDP no_sync pattern): grads
+        accumulate locally, no all-reduce fires."""
+        self._armed = sync
         for b in self.buckets:
             b.reset()
 
     def finalize(self):
-        if not self.enabled:
+        if not self.enabled or not self._armed:
             return
+        self._armed = False
         inv = 1.0 / self.world
         for b in self.buckets:
             if b.work is None:

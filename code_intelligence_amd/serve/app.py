@@ -58,15 +58,43 @@ def create_app(wrapper: InferenceWrapper | None = None,
     def healthz():
         return jsonify({"ok": True})
 
+    try:
+        from prometheus_client import Counter, Histogram, generate_latest
+        # re-registration (tests build several apps per process) -> reuse
+        try:
+            req_count = Counter("embedding_requests_total", "POST /text requests")
+            req_lat = Histogram("embedding_request_seconds", "request latency")
+        except ValueError:
+            from prometheus_client import REGISTRY
+            req_count = REGISTRY._names_to_collectors["embedding_requests_total"]
+            req_lat = REGISTRY._names_to_collectors["embedding_request_seconds"]
+    except ImportError:  # pragma: no cover
+        req_count = req_lat = generate_latest = None
+
     @app.route("/text", methods=["POST"])
     def text():
+        import time as _time
+        t0 = _time.perf_counter()
         data = request.get_json(force=True)
         doc = wrapper.process_dict({"title": data.get("title", ""),
                                     "body": data.get("body", "")})
         emb = wrapper.get_pooled_features(doc["text"]).numpy().astype("<f4")
         payload = emb.tobytes()
-        log.debug("embedding md5=%s", hashlib.md5(payload).hexdigest())
-        return payload, 200, {"Content-Type": "application/octet-stream"}
+        md5 = hashlib.md5(payload).hexdigest()
+        log.debug("embedding md5=%s", md5)
+        if req_count is not None:
+            req_count.inc()
+            req_lat.observe(_time.perf_counter() - t0)
+        # md5 echoed for the reference's change-debugging workflow
+        # (app.py:73-75 / repo_specific_model.py:179-181)
+        return payload, 200, {"Content-Type": "application/octet-stream",
+                              "X-Embedding-MD5": md5}
+
+    @app.route("/metrics")
+    def metrics():
+        if generate_latest is None:
+            return "prometheus_client not installed", 501
+        return generate_latest(), 200, {"Content-Type": "text/plain"}
 
     @app.route("/texts", methods=["POST"])
     def texts():

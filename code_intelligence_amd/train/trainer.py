@@ -82,17 +82,35 @@ class LMTrainer:
                         (r[:, 1:] - r[:, :-1]).pow(2).mean(dtype=torch.float32))
         return loss
 
-    def train_step(self, x, y, lr: float, mom: Optional[float] = None) -> float:
+    def train_step(self, x, y, lr: float, mom: Optional[float] = None,
+                   micro_batches: Optional[list] = None) -> float:
+        """One optimizer step. ``micro_batches`` (list of (x, y)) enables
+        gradient accumulation: losses averaged, all-reduce only on the
+        final micro-step (DDP no_sync)."""
         for g in self.opt.param_groups:
             g["lr"] = lr * self.lr_scale
             if mom is not None:
                 g["betas"] = (mom, g["betas"][1])
-        if self.dist is not None:
-            self.dist.prepare()
         self.opt.zero_grad(set_to_none=True)
-        loss = self.loss_on_batch(x, y)
-        loss.backward()
-        if x.is_cuda:
+        if micro_batches:
+            n = len(micro_batches)
+            total = 0.0
+            for i, (mx, my) in enumerate(micro_batches):
+                last = i == n - 1
+                if self.dist is not None:
+                    self.dist.prepare(sync=last)
+                loss = self.loss_on_batch(mx, my) / n
+                loss.backward()
+                total += float(loss.detach())
+            loss = total  # reported value: mean loss over micro-batches
+        else:
+            if self.dist is not None:
+                self.dist.prepare()
+            loss = self.loss_on_batch(x, y)
+            loss.backward()
+        ref_x = x if x is not None else (micro_batches[0][0]
+                                         if micro_batches else None)
+        if ref_x is not None and ref_x.is_cuda:
             from ..ops.lstm import sync_dw_stream
             sync_dw_stream()  # side-stream dW grads (CI_SIDE_DW) ordered
         if self.dist is not None:
@@ -101,7 +119,7 @@ class LMTrainer:
             torch.nn.utils.clip_grad_norm_(self._param_groups(), self.cfg.clip)
         self.opt.step()
         self.global_step += 1
-        return float(loss.detach())
+        return float(loss.detach()) if torch.is_tensor(loss) else float(loss)
 
     @torch.no_grad()
     def evaluate(self, loader: Iterable) -> dict:

@@ -81,3 +81,52 @@ def test_ddp_grads_match_single_process_average(tmp_path):
     assert len(shared) > 5
     for n in shared:
         assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
+
+
+def _worker_accum(rank, out_path):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(WORLD),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29517",
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        m = _build_model()
+        tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
+        mbs = [_make_batches(seed=200 + rank * 10 + i) for i in range(3)]
+        tr.train_step(None, None, lr=0.0, micro_batches=mbs)
+        if rank == 0:
+            grads = {n: p.grad.clone() for n, p in m.named_parameters()
+                     if p.grad is not None}
+            torch.save(grads, out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_gradient_accumulation_no_sync(tmp_path):
+    """3 micro-batches per rank, all-reduce only on the last: equals the
+    single-process mean over all 6 micro-batches."""
+    ctx = mp.get_context("spawn")
+    out = str(tmp_path / "g.pt")
+    procs = [ctx.Process(target=_worker_accum, args=(r, out)) for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=100)
+        assert p.exitcode == 0
+    dist_grads = torch.load(out, weights_only=True)
+
+    m = _build_model()
+    tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
+    acc = {}
+    for rank in range(WORLD):
+        for i in range(3):
+            for p_ in m.parameters():
+                p_.grad = None
+            m.reset()
+            x, y = _make_batches(seed=200 + rank * 10 + i)
+            (tr.loss_on_batch(x, y) / 3).backward()
+            for n, p_ in m.named_parameters():
+                if p_.grad is not None:
+                    acc[n] = acc.get(n, 0) + p_.grad / WORLD
+    for n in set(acc) & set(dist_grads):
+        assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
