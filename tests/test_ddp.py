@@ -119,10 +119,10 @@ def test_gradient_accumulation_no_sync(tmp_path):
     tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
     acc = {}
     for rank in range(WORLD):
+        m.reset()  # hidden state carries WITHIN a rank's micro-batches
         for i in range(3):
             for p_ in m.parameters():
                 p_.grad = None
-            m.reset()
             x, y = _make_batches(seed=200 + rank * 10 + i)
             (tr.loss_on_batch(x, y) / 3).backward()
             for n, p_ in m.named_parameters():
