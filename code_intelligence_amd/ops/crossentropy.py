@@ -144,6 +144,24 @@ class _FusedCEFunction(torch.autograd.Function):
                 db.to(weight.dtype) if has_bias else None, None)
 
 
+@torch.no_grad()
+def tied_decoder_accuracy(h: Tensor, weight: Tensor, bias: Tensor | None,
+                          targets: Tensor, chunk: int = 16384) -> Tensor:
+    """Top-1 next-token accuracy (eval metric; fastai reports accuracy
+    during fit). Chunked so the (N, 60k) logits never fully materialize."""
+    h2 = h.reshape(-1, h.shape[-1])
+    t2 = targets.reshape(-1)
+    w_t = weight.t()
+    correct = torch.zeros((), dtype=torch.long, device=h.device)
+    for s in range(0, h2.shape[0], chunk):
+        e = min(h2.shape[0], s + chunk)
+        logits = torch.mm(h2[s:e], w_t)
+        if bias is not None:
+            logits += bias
+        correct += (logits.argmax(dim=1) == t2[s:e]).sum()
+    return correct.float() / max(1, h2.shape[0])
+
+
 def tied_decoder_ce(h: Tensor, weight: Tensor, bias: Tensor | None,
                     targets: Tensor) -> Tensor:
     """h: (N, H) decoder input (already output-dropped); weight: (V, H) tied

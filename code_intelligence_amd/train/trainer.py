@@ -16,7 +16,7 @@ import torch
 from torch import nn
 
 from ..models.awd_lstm import AWDLSTM
-from ..ops.crossentropy import tied_decoder_ce
+from ..ops.crossentropy import tied_decoder_accuracy, tied_decoder_ce
 from ..ops.adam import FusedAdamW
 from ..parallel.ddp import DistributedGrads
 from .callbacks import Callback, CallbackList
@@ -122,18 +122,29 @@ class LMTrainer:
         return float(loss.detach()) if torch.is_tensor(loss) else float(loss)
 
     @torch.no_grad()
-    def evaluate(self, loader: Iterable) -> dict:
+    def evaluate(self, loader: Iterable, with_accuracy: bool = True) -> dict:
         self.model.eval()
         self.model.reset()
-        tot, n = 0.0, 0
+        tot, n, correct = 0.0, 0, 0.0
+        dec = self.model.decoder
         for x, y in loader:
-            loss = self.loss_on_batch(x, y)
+            enc_raw, enc_out = self.model.encoder(x)
+            out = dec.output_dp(enc_out[-1])
+            loss = tied_decoder_ce(out.reshape(-1, out.shape[-1]),
+                                   dec.decoder.weight, dec.decoder.bias,
+                                   y.reshape(-1))
             tot += float(loss) * x.numel()
+            if with_accuracy:
+                correct += float(tied_decoder_accuracy(
+                    out, dec.decoder.weight, dec.decoder.bias, y)) * x.numel()
             n += x.numel()
         self.model.train()
         self.model.reset()
         vl = tot / max(n, 1)
-        return {"valid_loss": vl, "valid_ppl": math.exp(min(vl, 30.0))}
+        metrics = {"valid_loss": vl, "valid_ppl": math.exp(min(vl, 30.0))}
+        if with_accuracy:
+            metrics["valid_acc"] = correct / max(n, 1)
+        return metrics
 
     # --- checkpoint/resume (full training state; the reference only
     # checkpoints model weights via fastai SaveModelCallback) -------------

@@ -61,3 +61,14 @@ def test_variational_dropout_locked_mask():
     # scaling preserves expectation roughly
     assert abs(y.mean().item() - 1.0) < 0.2
     assert torch.equal(variational_dropout(x, 0.5, training=False), x)
+
+
+def test_tied_decoder_accuracy():
+    from code_intelligence_amd.ops.crossentropy import tied_decoder_accuracy
+    torch.manual_seed(0)
+    w = torch.eye(6)  # logits = h -> argmax = largest coordinate
+    h = torch.zeros(4, 6)
+    h[0, 2] = 1; h[1, 5] = 1; h[2, 0] = 1; h[3, 1] = 1
+    t = torch.tensor([2, 5, 0, 3])  # 3 of 4 correct
+    acc = tied_decoder_accuracy(h, w, None, t, chunk=3)
+    assert float(acc) == 0.75
