@@ -179,3 +179,36 @@ def test_side_dw_grads_match():
     for n in base:
         assert torch.allclose(base[n], side[n], atol=1e-5), \
             (n, (base[n] - side[n]).abs().max())
+
+
+def test_full_pipeline_train_save_serve(tmp_path):
+    """GPU end-to-end: tokenized corpus -> one-cycle training -> artifact
+    save -> InferenceWrapper reload -> embeddings served (native kernels
+    throughout)."""
+    from code_intelligence_amd.data.lm_loader import LMStreamLoader
+    from code_intelligence_amd.data.synthetic import synthetic_issue_tokens
+    from code_intelligence_amd.engine.inference import (InferenceWrapper,
+                                                        save_artifacts)
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+    from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
+
+    torch.manual_seed(0)
+    vocab_sz = 2000
+    docs = synthetic_issue_tokens(200, vocab_sz, markov=True, seed=2)
+    model = AWDLSTM(vocab_sz=vocab_sz, emb_sz=128, n_hid=256, n_layers=2) \
+        .to(DEV, torch.bfloat16)
+    tr = LMTrainer(model, TrainConfig())
+    dl = LMStreamLoader(docs, bs=16, bptt=48, device=torch.device(DEV))
+    m0 = tr.evaluate(dl, with_accuracy=True)
+    tr.fit(dl, epochs=2)
+    m1 = tr.evaluate(dl, with_accuracy=True)
+    assert m1["valid_loss"] < m0["valid_loss"]
+
+    vocab = Vocab(defaults_specials + [f"w{i}" for i in range(vocab_sz - 9)])
+    save_artifacts(model, vocab, tmp_path / "art")
+    w = InferenceWrapper(model_path=str(tmp_path / "art"), device=DEV)
+    out = w.texts_to_embedding(["w3 w5 w9", "w1 " * 40], bs=2)
+    assert out.shape == (2, 3 * 128)
+    import numpy as np
+    assert np.isfinite(out).all()
