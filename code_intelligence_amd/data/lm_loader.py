@@ -22,6 +22,8 @@ class LMStreamLoader:
     def __init__(self, docs: List[List[int]], bs: int, bptt: int,
                  bos_idx: Optional[int] = 2, shuffle: bool = True,
                  seed: int = 0, device: Optional[torch.device] = None):
+        if bs < 1 or bptt < 1:
+            raise ValueError(f"bs and bptt must be >= 1 (got {bs}, {bptt})")
         self.docs, self.bs, self.bptt = docs, bs, bptt
         self.bos_idx, self.shuffle, self.seed = bos_idx, shuffle, seed
         self.device = device

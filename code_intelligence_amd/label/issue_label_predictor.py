@@ -63,10 +63,14 @@ class IssueLabelPredictor:
         kind = spec.get("kind")
         org = spec.get("org")
         repo = spec.get("repo")
+        if not org:
+            raise ValueError(f"model spec needs 'org': {spec}")
         if kind == "automl":
             m: IssueLabelModel = AutoMLModel(
                 model_name=spec["model"], endpoint=spec.get("endpoint"))
         elif kind == "repo_specific":
+            if not repo:
+                raise ValueError(f"repo_specific spec needs 'repo': {spec}")
             m = RepoSpecificLabelModel.from_repo(
                 org, repo, embedding_api_endpoint=self.embedding_api_endpoint)
         else:

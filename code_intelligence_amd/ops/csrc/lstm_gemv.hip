@@ -97,6 +97,7 @@ void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16,
               "gemv cell kernel is bf16");
   const int T = xp.size(0), B = xp.size(1);
+  TORCH_CHECK(B <= 8, "gemv cell kernel is for B <= 8 (got ", B, ")");
   const int H = w_hh.size(1);
   auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
   auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
