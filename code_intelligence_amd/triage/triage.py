@@ -139,7 +139,7 @@ query issues($org: String!, $repo: String!, $pageSize: Int!, $cursor: String,
       totalCount
       pageInfo { endCursor hasNextPage }
       edges { node {
-        number title url state closedAt
+        id number title url state closedAt
         labels(first: 30) { edges { node { name } } }
         projectCards(first: 30) {
           edges { node { id project { name number } } } }
@@ -160,7 +160,7 @@ ISSUE_QUERY = """
 query issue($url: URI!) {
   resource(url: $url) {
     ... on Issue {
-      number title url state closedAt
+      id number title url state closedAt
       labels(first: 100) { edges { node { name } } }
       projectCards(first: 100) { edges { node { id project { name number } } } }
       timelineItems(first: 100, itemTypes: [LABELED_EVENT,
@@ -265,9 +265,13 @@ class IssueTriage:
         return result
 
     def _add_triage_project(self, info: TriageInfo) -> None:
+        content_id = info.issue.get("id")
+        if content_id is None:
+            raise ValueError(
+                "issue node id missing (the triage queries request it); "
+                "cannot build the addProjectCard mutation")
         self.client.run_query(ADD_CARD_MUTATION, {
-            "cardId": self.project_card_id,
-            "contentId": info.issue.get("id") or info.issue.get("number")})
+            "cardId": self.project_card_id, "contentId": content_id})
 
     def _remove_triage_project(self, info: TriageInfo) -> None:
         self.client.run_query(DELETE_CARD_MUTATION, {

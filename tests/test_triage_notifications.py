@@ -17,7 +17,8 @@ def _edges(items):
 
 def _issue(state="OPEN", labels=(), events=(), cards=(), closed_at=None):
     return {
-        "number": 1, "url": "https://github.com/a/b/issues/1", "state": state,
+        "id": "MDU6SXNzdWUx", "number": 1,
+        "url": "https://github.com/a/b/issues/1", "state": state,
         "closedAt": closed_at,
         "labels": _edges([{"name": n} for n in labels]),
         "projectCards": _edges(list(cards)),
@@ -102,6 +103,7 @@ def test_triage_adds_and_removes_cards():
                                               "removed_from_project"]
     assert len(client.mutations) == 2
     assert "addCard" in client.mutations[0][0]
+    assert client.mutations[0][1]["contentId"] == "MDU6SXNzdWUx"
     assert client.mutations[1][1] == {"cardId": "card1"}
 
 
