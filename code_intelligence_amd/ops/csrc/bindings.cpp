@@ -13,6 +13,9 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
 void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
                            at::Tensor cs, at::Tensor gates);
+void lstm_seq_forward_gemv_fp8(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                               at::Tensor c0, at::Tensor w8, at::Tensor wscale,
+                               at::Tensor hs, at::Tensor cs, at::Tensor gates);
 void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
                        at::Tensor gates, at::Tensor hs, at::Tensor cs,
                        at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
@@ -41,6 +44,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "LSTM sequence forward (fused MFMA cell kernel)");
   m.def("lstm_seq_forward_gemv", &ci::lstm_seq_forward_gemv,
         "LSTM sequence forward (fused GEMV+cell kernel, small batch)");
+  m.def("lstm_seq_forward_gemv_fp8", &ci::lstm_seq_forward_gemv_fp8,
+        "LSTM sequence forward (fp8-weight GEMV+cell kernel)");
   m.def("lstm_seq_backward", &ci::lstm_seq_backward,
         "LSTM sequence backward (pointwise + per-step GEMM)");
   m.def("concat_pool", &ci::concat_pool, "masked mean/max/last concat pool");

@@ -118,4 +118,126 @@ void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   }
 }
 
+
+// ---- fp8 (OCP e4m3) weight variant --------------------------------------
+// Serving is bound by streaming W_hh (46 MB/layer-step at the deployed
+// shape); storing W as e4m3 with per-row scales halves that stream.
+// Hardware unpack: __builtin_amdgcn_cvt_pk_f32_fp8 converts packed fp8
+// pairs at VALU rate. Opt-in (CI_SERVE_FP8W=1), eval-path only.
+namespace {
+__device__ __forceinline__ void fp8x4_to_f32(unsigned int u, float* out) {
+  typedef float f32x2 __attribute__((ext_vector_type(2)));
+  f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(u, false);
+  f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(u, true);
+  out[0] = lo[0]; out[1] = lo[1]; out[2] = hi[0]; out[3] = hi[1];
+}
+}  // namespace
+
+__global__ __launch_bounds__(256) void lstm_cell_gemv_fp8(
+    const __hip_bfloat16* __restrict__ h_prev, long h_rs,
+    const unsigned char* __restrict__ w8,      // (4H, H) e4m3
+    const float* __restrict__ wscale,          // (4H) per-row scales
+    const __hip_bfloat16* __restrict__ xp, long xp_rs,
+    const float* __restrict__ bias,
+    const float* __restrict__ c_prev, long cp_rs,
+    __hip_bfloat16* __restrict__ h_out, long ho_rs,
+    float* __restrict__ c_out, long co_rs,
+    __hip_bfloat16* __restrict__ gates_out, long go_rs,
+    int B, int H) {
+  constexpr int BMAX = 8;
+  const int j = blockIdx.x;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int row = wave * H + j;
+  const unsigned char* wrow = w8 + (long)row * H;
+  __shared__ float dots[BMAX][4];
+  const int Hv = H / 16 * 16;
+  float acc[BMAX];
+  #pragma unroll
+  for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
+  float wf[16];
+  for (int k = lane * 16; k < Hv; k += 64 * 16) {
+    const uint4 wq = *reinterpret_cast<const uint4*>(wrow + k);  // 16 fp8
+    fp8x4_to_f32(wq.x, wf);
+    fp8x4_to_f32(wq.y, wf + 4);
+    fp8x4_to_f32(wq.z, wf + 8);
+    fp8x4_to_f32(wq.w, wf + 12);
+    #pragma unroll
+    for (int b = 0; b < BMAX; ++b) {
+      if (b >= B) break;
+      const bf16x8g h0v = *reinterpret_cast<const bf16x8g*>(
+          h_prev + (long)b * h_rs + k);
+      const bf16x8g h1v = *reinterpret_cast<const bf16x8g*>(
+          h_prev + (long)b * h_rs + k + 8);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) acc[b] += wf[e] * (float)h0v[e];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) acc[b] += wf[8 + e] * (float)h1v[e];
+    }
+  }
+  for (int k = Hv + lane; k < H; k += 64) {
+    float w1[4];
+    fp8x4_to_f32((unsigned int)wrow[k], w1);  // low byte -> w1[0]
+    #pragma unroll
+    for (int b = 0; b < BMAX; ++b) {
+      if (b >= B) break;
+      acc[b] += w1[0] * ld(h_prev + (long)b * h_rs + k);
+    }
+  }
+  const float sc = wscale[row];
+  #pragma unroll
+  for (int b = 0; b < BMAX; ++b) {
+    if (b >= B) break;
+    float a = acc[b] * sc;
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      a += __shfl_down(a, off);
+    if (lane == 0) dots[b][wave] = a;
+  }
+  __syncthreads();
+  for (int b = threadIdx.x; b < B; b += blockDim.x) {
+    const long xo = (long)b * xp_rs + j;
+    const float gi = sigmoidf_(dots[b][0] + ld(xp + xo) + bias[j]);
+    const float gf = sigmoidf_(dots[b][1] + ld(xp + xo + H) + bias[j + H]);
+    const float gg = tanhf(dots[b][2] + ld(xp + xo + 2 * H) + bias[j + 2 * H]);
+    const float go = sigmoidf_(dots[b][3] + ld(xp + xo + 3 * H) + bias[j + 3 * H]);
+    const float c = gf * c_prev[(long)b * cp_rs + j] + gi * gg;
+    const float h = go * tanhf(c);
+    st(h_out + (long)b * ho_rs + j, h);
+    c_out[(long)b * co_rs + j] = c;
+    const long g0 = (long)b * go_rs + j;
+    st(gates_out + g0, gi);
+    st(gates_out + g0 + H, gf);
+    st(gates_out + g0 + 2 * H, gg);
+    st(gates_out + g0 + 3 * H, go);
+  }
+}
+
+void lstm_seq_forward_gemv_fp8(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                               at::Tensor c0, at::Tensor w8, at::Tensor wscale,
+                               at::Tensor hs, at::Tensor cs, at::Tensor gates) {
+  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
+  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates); CI_CHECK_CONTIG(w8);
+  const int T = xp.size(0), B = xp.size(1);
+  TORCH_CHECK(B <= 8, "fp8 gemv kernel is for B <= 8");
+  const int H = w8.size(1);
+  auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
+  auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
+  auto* gp = reinterpret_cast<__hip_bfloat16*>(gates.data_ptr());
+  auto h0c = h0.contiguous();
+  auto* h0p = reinterpret_cast<const __hip_bfloat16*>(h0c.data_ptr());
+  for (int t = 0; t < T; ++t) {
+    const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * B * H;
+    const float* cp = (t == 0) ? c0.data_ptr<float>()
+                               : cs.data_ptr<float>() + (long)(t - 1) * B * H;
+    hipLaunchKernelGGL(lstm_cell_gemv_fp8, dim3(H), dim3(256), 0, stream(),
+        hp, (long)H, w8.data_ptr<unsigned char>(), wscale.data_ptr<float>(),
+        xpp + (long)t * B * 4 * H, (long)4 * H,
+        bias.data_ptr<float>(), cp, (long)H,
+        hsp + (long)t * B * H, (long)H,
+        cs.data_ptr<float>() + (long)t * B * H, (long)H,
+        gp + (long)t * B * 4 * H, (long)4 * H, B, H);
+  }
+}
+
 }  // namespace ci

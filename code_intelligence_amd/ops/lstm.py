@@ -73,6 +73,25 @@ def _accum_grad(p: Tensor, g: Tensor) -> None:
         p.grad.add_(g.to(p.dtype))
 
 
+_fp8_cache: dict = {}
+
+
+def _fp8_weights(w_hh: Tensor):
+    """Per-row e4m3 quantization of the recurrent weight, cached by tensor
+    identity+version (serving weights are static). CI_SERVE_FP8W=1 halves
+    the weight stream that bounds small-batch serve latency."""
+    key = (id(w_hh), w_hh._version)
+    hit = _fp8_cache.get(key)
+    if hit is not None:
+        return hit
+    scale = w_hh.detach().abs().amax(dim=1, keepdim=True).float()         .clamp_min(1e-12) / 448.0
+    q = (w_hh.detach().float() / scale).clamp(-448.0, 448.0)         .to(torch.float8_e4m3fn).view(torch.uint8).contiguous()
+    out = (q, scale.squeeze(1).contiguous())
+    _fp8_cache.clear()  # keep at most the live weight set small
+    _fp8_cache[key] = out
+    return out
+
+
 def _cpu_lstm_loop(x: Tensor, h0: Tensor, c0: Tensor, w_ih: Tensor, w_hh: Tensor,
                    b_ih: Tensor, b_hh: Tensor) -> Tuple[Tensor, Tensor, Tensor]:
     """Pure-PyTorch reference (autograd-capable). x: (B,T,In)."""
@@ -115,8 +134,14 @@ class _FusedLSTMFunction(torch.autograd.Function):
         if dt == torch.bfloat16 and B <= 8 and mode != "lib-only":
             # serve/decode regime: weight-streaming-bound GEMV — the fused
             # GEMV+cell kernel replaces hipBLASLt GEMV + pointwise launch
-            lib.lstm_seq_forward_gemv(xp, bias, h0, c0.to(torch.float32),
-                                      w_hh, hs, cs, gates)
+            if os.environ.get("CI_SERVE_FP8W", "0") == "1"                     and not torch.is_grad_enabled():
+                w8, wscale = _fp8_weights(w_hh)
+                lib.lstm_seq_forward_gemv_fp8(xp, bias, h0,
+                                              c0.to(torch.float32), w8,
+                                              wscale, hs, cs, gates)
+            else:
+                lib.lstm_seq_forward_gemv(xp, bias, h0, c0.to(torch.float32),
+                                          w_hh, hs, cs, gates)
         elif mode == "fused" and dt == torch.bfloat16 and H % 8 == 0:
             lib.lstm_seq_forward_fused(xp, bias, h0, c0.to(torch.float32), w_hh,
                                        hs, cs, gates)

@@ -212,3 +212,25 @@ def test_full_pipeline_train_save_serve(tmp_path):
     assert out.shape == (2, 3 * 128)
     import numpy as np
     assert np.isfinite(out).all()
+
+
+def test_fp8_weight_serving_close_to_bf16():
+    """CI_SERVE_FP8W=1: fp8-weight GEMV embeddings stay close to the bf16
+    path (per-row scales; serving-only opt-in)."""
+    from code_intelligence_amd.engine.inference import InferenceWrapper
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+    torch.manual_seed(0)
+    vocab = Vocab(defaults_specials + [f"w{i}" for i in range(400)])
+    m = AWDLSTM(vocab_sz=len(vocab), emb_sz=64, n_hid=128, n_layers=2)
+    w = InferenceWrapper(encoder=m.encoder, vocab=vocab, device=DEV)
+    text = "w3 w17 w5 " * 20
+    base = w.get_pooled_features(text)
+    os.environ["CI_SERVE_FP8W"] = "1"
+    try:
+        fp8 = w.get_pooled_features(text)
+    finally:
+        os.environ.pop("CI_SERVE_FP8W", None)
+    cos = torch.nn.functional.cosine_similarity(base, fp8).item()
+    assert cos > 0.995, cos
+    assert (base - fp8).abs().max() < 0.1, (base - fp8).abs().max()
