@@ -344,3 +344,44 @@ def test_feedback_collector(tmp_path):
     assert records[0]["labels"] == ["kind/bug", "area/ops"]
     assert records[0]["score"] == 2
     assert out.read_text().count("\n") == 1
+
+
+def test_github_issue_client_rest_contract():
+    """GitHubIssueClient hits the documented REST endpoints with the
+    token header (worker's GitHub boundary)."""
+    from code_intelligence_amd.label.worker import GitHubIssueClient
+
+    calls = []
+
+    class Sess:
+        def post(self, url, json=None, headers=None):
+            calls.append(("POST", url, json, headers))
+            class R:
+                def raise_for_status(self):
+                    pass
+                def json(self):
+                    return {}
+            return R()
+
+        def get(self, url, headers=None):
+            calls.append(("GET", url, None, headers))
+            class R:
+                def raise_for_status(self):
+                    pass
+                def json(self):
+                    return [{"body": "x"}]
+            return R()
+
+    class Tok:
+        def auth_headers(self):
+            return {"Authorization": "token sekrit"}
+
+    c = GitHubIssueClient(token_generator=Tok(), session=Sess())
+    c.add_labels("o", "r", 5, ["kind/bug"])
+    c.add_comment("o", "r", 5, "hello")
+    assert c.list_comments("o", "r", 5) == [{"body": "x"}]
+    assert calls[0][1].endswith("/repos/o/r/issues/5/labels")
+    assert calls[0][2] == {"labels": ["kind/bug"]}
+    assert calls[0][3]["Authorization"] == "token sekrit"
+    assert calls[1][1].endswith("/repos/o/r/issues/5/comments")
+    assert calls[2][0] == "GET"
