@@ -58,8 +58,10 @@ class MLPWrapper:
     # --- training ---------------------------------------------------------
     def fit(self, X: np.ndarray, y: np.ndarray, epochs: Optional[int] = None,
             batch_size: int = 200, verbose: bool = False) -> "MLPWrapper":
-        X_t = torch.as_tensor(np.asarray(X), dtype=torch.float32, device=self.device)
-        y_t = torch.as_tensor(np.asarray(y), dtype=torch.float32, device=self.device)
+        X_t = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32,
+                              device=self.device)
+        y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32,
+                              device=self.device)
         if y_t.dim() == 1:
             y_t = y_t.unsqueeze(1)
         self.clf = self.clf.to(self.device)
@@ -108,7 +110,7 @@ class MLPWrapper:
                                     ) -> Dict[int, Optional[float]]:
         from sklearn.metrics import precision_recall_curve
         probs = self.predict_probabilities(X)
-        y = np.asarray(y)
+        y = np.asarray(y).copy()  # torch.as_tensor needs writable arrays
         if y.ndim == 1:
             y = y[:, None]
         out: Dict[int, Optional[float]] = {}

@@ -173,5 +173,5 @@ def train_universal_model(org: str, archive_root=None, epochs: int = 5,
     net.eval()
     model = UniversalKindLabelModel(net.cpu(), vocab, prefix=prefix)
     log.info("trained universal model on %d issues (final loss %.4f)",
-             len(X), float(loss))
+             len(X), float(loss.detach()))
     return model
