@@ -98,7 +98,8 @@ class MLPWrapper:
         return self
 
     def predict_probabilities(self, X: np.ndarray) -> np.ndarray:
-        X_t = torch.as_tensor(np.asarray(X), dtype=torch.float32, device=self.device)
+        X_t = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32,
+                              device=self.device)
         with torch.no_grad():
             p = torch.sigmoid(self.clf.to(self.device)(X_t))
         return p.cpu().numpy()
