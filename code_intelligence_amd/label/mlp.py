@@ -21,6 +21,15 @@ import torch
 from torch import nn
 
 
+def _to_t(a, device) -> torch.Tensor:
+    """float32 tensor from array-like; copies only when non-writable
+    (np.load/npz arrays are read-only and trip torch.as_tensor)."""
+    arr = np.asarray(a, dtype=np.float32)
+    if not arr.flags.writeable:
+        arr = arr.copy()
+    return torch.as_tensor(arr, device=device)
+
+
 class MLPHead(nn.Module):
     """input (1600-d truncated embedding) -> hidden(600) -> hidden(600) ->
     sigmoid multi-label output (repo_mlp.ipynb cell 28 shape)."""
@@ -58,10 +67,8 @@ class MLPWrapper:
     # --- training ---------------------------------------------------------
     def fit(self, X: np.ndarray, y: np.ndarray, epochs: Optional[int] = None,
             batch_size: int = 200, verbose: bool = False) -> "MLPWrapper":
-        X_t = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32,
-                              device=self.device)
-        y_t = torch.as_tensor(np.ascontiguousarray(y), dtype=torch.float32,
-                              device=self.device)
+        X_t = _to_t(X, self.device)
+        y_t = _to_t(y, self.device)
         if y_t.dim() == 1:
             y_t = y_t.unsqueeze(1)
         self.clf = self.clf.to(self.device)
@@ -98,8 +105,7 @@ class MLPWrapper:
         return self
 
     def predict_probabilities(self, X: np.ndarray) -> np.ndarray:
-        X_t = torch.as_tensor(np.ascontiguousarray(X), dtype=torch.float32,
-                              device=self.device)
+        X_t = _to_t(X, self.device)
         with torch.no_grad():
             p = torch.sigmoid(self.clf.to(self.device)(X_t))
         return p.cpu().numpy()
