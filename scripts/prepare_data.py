@@ -79,6 +79,11 @@ def main():
     torch.save(docs, out / "docs.pt")
     vocab.save(out / "vocab.json")
     n_tok = sum(len(d) for d in docs)
+    # corpus stats (the counts the reference's 01 notebook reports)
+    (out / "corpus_stats.json").write_text(json.dumps({
+        "n_docs": len(docs), "n_tokens": n_tok, "vocab_size": len(vocab),
+        "mean_doc_tokens": round(n_tok / max(1, len(docs)), 1),
+        "max_vocab": args.max_vocab, "min_freq": args.min_freq}))
     print(f"vocab {len(vocab)}; {n_tok} tokens -> {out}")
 
 
