@@ -140,3 +140,57 @@ def test_download_issues_shards(tmp_path):
     got = download_issues("kubeflow/kubeflow", tmp_path, client=client)
     assert len(got) == 5
     assert len(parse_issue_shards(tmp_path)) == 5
+
+
+def test_mark_read_dry_run_and_patch():
+    class Sess:
+        def __init__(self):
+            self.patched = []
+            self.page = 0
+
+        def get(self, url, params=None, headers=None):
+            self.page += 1
+            batch = [{"id": "1", "reason": "subscribed"},
+                     {"id": "2", "reason": "mention"}] if self.page == 1 else []
+
+            class R:
+                status_code = 200
+
+                def raise_for_status(self):
+                    pass
+
+                def json(self2):
+                    return batch
+            return R()
+
+        def patch(self, url, headers=None):
+            self.patched.append(url)
+
+            class R:
+                status_code = 205
+            return R()
+
+    s = Sess()
+    nm = NotificationManager(session=s, token="t")
+    marked = nm.mark_read(dry_run=True)
+    assert marked == ["1"]            # mention kept, nothing patched
+    assert s.patched == []
+    s.page = 0
+    marked = nm.mark_read()
+    assert marked == ["1"]
+    assert s.patched and s.patched[0].endswith("/notifications/threads/1")
+
+
+def test_archive_max_age_filter(tmp_path):
+    import pandas as pd
+    from code_intelligence_amd.gh import bigquery
+    now = pd.Timestamp.now(tz="UTC")
+    events = [
+        {"org": "o", "repo": "r", "issue_num": 1, "title": "new", "body": "",
+         "labels": [], "updated_at": (now - pd.Timedelta(days=2)).isoformat()},
+        {"org": "o", "repo": "r", "issue_num": 2, "title": "old", "body": "",
+         "labels": [], "updated_at": (now - pd.Timedelta(days=90)).isoformat()},
+    ]
+    bigquery.write_archive_events(events, tmp_path / "e.jsonl")
+    df = bigquery.get_issues("o", max_age_days=30, archive_root=tmp_path)
+    assert df["issue_num"].tolist() == [1]
