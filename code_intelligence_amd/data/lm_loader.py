@@ -26,7 +26,7 @@ class LMStreamLoader:
             raise ValueError(f"bs and bptt must be >= 1 (got {bs}, {bptt})")
         self.docs, self.bs, self.bptt = docs, bs, bptt
         self.bos_idx, self.shuffle, self.seed = bos_idx, shuffle, seed
-        self.device = device
+        self.device = torch.device(device) if device is not None else None
         self.epoch = 0
 
     def _stream(self) -> Tensor:
@@ -54,6 +54,11 @@ class LMStreamLoader:
         if per < 2:
             return
         mat = stream[: per * self.bs].view(self.bs, per)
+        if self.device is not None and self.device.type == "cuda" \
+                and torch.cuda.is_available():
+            # one pinned epoch matrix => every window's .to(non_blocking=True)
+            # is a real async H2D copy instead of a silent sync one
+            mat = mat.contiguous().pin_memory()
         n_batches = (per - 1) // self.bptt
         for k in range(n_batches):
             s = k * self.bptt
