@@ -1,8 +1,9 @@
 from .awd_lstm import (AWDLSTM, AWDLSTMEncoder, EmbeddingDropout,
                        LinearDecoder, RNNDropout, WeightDroppedLSTM,
-                       awd_lstm_lm_config)
+                       WeightDroppedQRNN, awd_lstm_lm_config)
 
 __all__ = [
     "AWDLSTM", "AWDLSTMEncoder", "EmbeddingDropout", "LinearDecoder",
-    "RNNDropout", "WeightDroppedLSTM", "awd_lstm_lm_config",
+    "RNNDropout", "WeightDroppedLSTM", "WeightDroppedQRNN",
+    "awd_lstm_lm_config",
 ]

@@ -31,6 +31,7 @@ from ..ops.dropout import variational_dropout
 __all__ = [
     "EmbeddingDropout",
     "WeightDroppedLSTM",
+    "WeightDroppedQRNN",
     "RNNDropout",
     "AWDLSTMEncoder",
     "LinearDecoder",
@@ -46,6 +47,8 @@ awd_lstm_lm_config = dict(
     n_hid=1150,
     n_layers=3,
     pad_token=1,
+    qrnn=False,
+    bidir=False,
     tie_weights=True,
     out_bias=True,
     output_p=0.1,
@@ -166,6 +169,48 @@ class WeightDroppedLSTM(nn.Module):
         super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
 
+class WeightDroppedQRNN(nn.Module):
+    """Single QRNN layer (reference ``--qrnn`` flag, train.py:43).
+
+    Gates for every timestep come from ONE hipBLASLt GEMM; the sequential
+    part is only the elementwise fo-pool scan (ops/qrnn.py — HIP kernel on
+    GPU). DropConnect (weight_p) applies to the gate projection like fastai
+    wraps QRNNLayer.linear. window=2 keeps the previous window's last x
+    (save_prev_x) so BPTT windows chain exactly.
+    """
+
+    def __init__(self, input_size: int, hidden_size: int,
+                 weight_p: float = 0.0, window: int = 1):
+        super().__init__()
+        from ..ops.qrnn import qrnn_forward
+        self._qrnn_forward = qrnn_forward
+        self.input_size, self.hidden_size = input_size, hidden_size
+        self.weight_p, self.window = weight_p, window
+        self.weight_raw = nn.Parameter(
+            torch.empty(3 * hidden_size, window * input_size))
+        self.bias = nn.Parameter(torch.zeros(3 * hidden_size))
+        self.prev_x: Optional[Tensor] = None
+        stdv = 1.0 / math.sqrt(hidden_size)
+        nn.init.uniform_(self.weight_raw, -stdv, stdv)
+
+    def reset(self) -> None:
+        self.prev_x = None
+
+    def forward(self, x: Tensor, state: Tuple[Tensor, Tensor]
+                ) -> Tuple[Tensor, Tuple[Tensor, Tensor]]:
+        w = nn.functional.dropout(self.weight_raw, p=self.weight_p,
+                                  training=self.training)
+        prev = self.prev_x
+        if prev is not None and (prev.size(0) != x.size(0)
+                                 or prev.device != x.device):
+            prev = None
+        h, cT = self._qrnn_forward(x, state[1], w, self.bias,
+                                   self.window, prev)
+        if self.window == 2:
+            self.prev_x = x[:, -1:].detach()
+        return h, (h[:, -1], cT)
+
+
 class AWDLSTMEncoder(nn.Module):
     """The encoder stack: embedding (+dropout) → n_layers weight-dropped LSTMs.
 
@@ -179,21 +224,34 @@ class AWDLSTMEncoder(nn.Module):
 
     def __init__(self, vocab_sz: int, emb_sz: int, n_hid: int, n_layers: int,
                  pad_token: int = 1, hidden_p: float = 0.15, input_p: float = 0.25,
-                 embed_p: float = 0.02, weight_p: float = 0.2):
+                 embed_p: float = 0.02, weight_p: float = 0.2,
+                 qrnn: bool = False):
         super().__init__()
         self.vocab_sz, self.emb_sz, self.n_hid, self.n_layers = vocab_sz, emb_sz, n_hid, n_layers
         self.pad_token = pad_token
+        self.qrnn = qrnn
         self.bs = 1
         self.encoder = nn.Embedding(vocab_sz, emb_sz, padding_idx=pad_token)
         self.encoder.weight.data.uniform_(-self.initrange, self.initrange)
         self.encoder_dp = EmbeddingDropout(self.encoder, embed_p)
-        self.rnns = nn.ModuleList([
-            WeightDroppedLSTM(
-                emb_sz if l == 0 else n_hid,
-                n_hid if l != n_layers - 1 else emb_sz,
-                weight_p=weight_p)
-            for l in range(n_layers)
-        ])
+        if qrnn:
+            # reference --qrnn flag (train.py:43); fastai windows: 2 on the
+            # first layer, 1 afterwards
+            self.rnns = nn.ModuleList([
+                WeightDroppedQRNN(
+                    emb_sz if l == 0 else n_hid,
+                    n_hid if l != n_layers - 1 else emb_sz,
+                    weight_p=weight_p, window=2 if l == 0 else 1)
+                for l in range(n_layers)
+            ])
+        else:
+            self.rnns = nn.ModuleList([
+                WeightDroppedLSTM(
+                    emb_sz if l == 0 else n_hid,
+                    n_hid if l != n_layers - 1 else emb_sz,
+                    weight_p=weight_p)
+                for l in range(n_layers)
+            ])
         self.input_dp = RNNDropout(input_p)
         self.hidden_dps = nn.ModuleList([RNNDropout(hidden_p) for _ in range(n_layers)])
         self.hidden: List[Tuple[Tensor, Tensor]] = []
@@ -209,6 +267,9 @@ class AWDLSTMEncoder(nn.Module):
         if bs is not None:
             self.bs = bs
         self.hidden = [self._one_hidden(l, self.bs) for l in range(self.n_layers)]
+        for rnn in self.rnns:
+            if hasattr(rnn, "reset"):
+                rnn.reset()  # QRNN: clear the saved window-2 prev_x
 
     def select_hidden(self, idxs: Tensor) -> None:
         self.hidden = [(h[idxs], c[idxs]) for h, c in self.hidden]
@@ -268,10 +329,16 @@ class AWDLSTM(nn.Module):
     def __init__(self, vocab_sz: int, emb_sz: int = 400, n_hid: int = 1150,
                  n_layers: int = 3, pad_token: int = 1, tie_weights: bool = True,
                  out_bias: bool = True, output_p: float = 0.1, hidden_p: float = 0.15,
-                 input_p: float = 0.25, embed_p: float = 0.02, weight_p: float = 0.2):
+                 input_p: float = 0.25, embed_p: float = 0.02, weight_p: float = 0.2,
+                 qrnn: bool = False, bidir: bool = False):
         super().__init__()
+        if bidir:
+            raise NotImplementedError(
+                "bidir=True is not supported: the reference exposes the flag "
+                "(train.py:43) but never enables it, and a bidirectional LM "
+                "leaks future tokens into next-token prediction")
         encoder = AWDLSTMEncoder(vocab_sz, emb_sz, n_hid, n_layers, pad_token,
-                                 hidden_p, input_p, embed_p, weight_p)
+                                 hidden_p, input_p, embed_p, weight_p, qrnn=qrnn)
         decoder = LinearDecoder(vocab_sz, emb_sz, output_p,
                                 tie_encoder=encoder.encoder if tie_weights else None,
                                 bias=out_bias)

@@ -21,6 +21,10 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
                        at::Tensor c0, at::Tensor w_hh, at::Tensor dgates,
                        at::Tensor dh0, at::Tensor dc0);
 at::Tensor concat_pool(at::Tensor hidden, at::Tensor lengths);
+std::vector<at::Tensor> qrnn_fo_pool_fwd(at::Tensor gates, at::Tensor c0);
+std::vector<at::Tensor> qrnn_fo_pool_bwd(at::Tensor gates, at::Tensor c,
+                                         at::Tensor c0, at::Tensor dh,
+                                         at::Tensor dcT);
 void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor bias,
                  at::Tensor lse, at::Tensor tgt);
 void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
@@ -49,6 +53,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_backward", &ci::lstm_seq_backward,
         "LSTM sequence backward (pointwise + per-step GEMM)");
   m.def("concat_pool", &ci::concat_pool, "masked mean/max/last concat pool");
+  m.def("qrnn_fo_pool_fwd", &ci::qrnn_fo_pool_fwd,
+        "QRNN fo-pool forward scan (activates gates in place)");
+  m.def("qrnn_fo_pool_bwd", &ci::qrnn_fo_pool_bwd,
+        "QRNN fo-pool backward scan (pre-activation gate grads)");
   m.def("ce_rowstats", &ci::ce_rowstats, "CE row logsumexp + target logit");
   m.def("ce_dlogits", &ci::ce_dlogits, "in-place (softmax-onehot)*scale");
   m.def("fused_adamw", &ci::fused_adamw, "fused AdamW step");

@@ -1,0 +1,131 @@
+// QRNN fo-pooling scan (the reference exposes --qrnn on its train CLI:
+// Issue_Embeddings/train.py:43, hyperparam_sweep/lm_tune.py:43,52-53 —
+// "requires CudNN" there; here it is a native CDNA4 kernel).
+//
+// QRNN shape: the 3-gate projection (z|f|o) is ONE T-parallel GEMM done by
+// hipBLASLt outside this file — the only sequential part is this elementwise
+// scan:  c_t = f_t * c_{t-1} + (1 - f_t) * z_t ,  h_t = o_t * c_t.
+//
+// Layout (B, T, 3H) gates / (B, T, H) states; one thread per (b, j) lane
+// looping over T: consecutive threads touch consecutive j => every load and
+// store in the t-loop is a fully-coalesced 256-thread row. B*H lanes
+// (deployed config: 104*2400 ≈ 250k threads ≈ 975 workgroups) fill the 256
+// CUs without any cross-thread dependency, so the scan runs at HBM speed.
+//
+// The forward kernel also APPLIES the activations (tanh/sigmoid/sigmoid) in
+// place over the GEMM output, saving a separate elementwise pass over the
+// (B,T,3H) buffer; backward emits pre-activation gate grads so dX/dW are
+// again single GEMMs.
+#include "common.h"
+
+namespace ci {
+
+template <typename T>
+__global__ void qrnn_fo_fwd_kernel(T* __restrict__ gates,   // (B,T,3H) preact -> act
+                                   const T* __restrict__ c0,  // (B,H)
+                                   T* __restrict__ h,          // (B,T,H)
+                                   T* __restrict__ c,          // (B,T,H)
+                                   int B, int Tn, int H) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * H) return;
+  const int b = idx / H, j = idx % H;
+  T* g = gates + (long)b * Tn * 3 * H + j;
+  T* hr = h + (long)b * Tn * H + j;
+  T* cr = c + (long)b * Tn * H + j;
+  float cv = ld(c0 + (long)b * H + j);
+  for (int t = 0; t < Tn; ++t) {
+    T* gt = g + (long)t * 3 * H;
+    const float z = tanhf(ld(gt));
+    const float f = sigmoidf_(ld(gt + H));
+    const float o = sigmoidf_(ld(gt + 2 * H));
+    st(gt, z); st(gt + H, f); st(gt + 2 * H, o);
+    cv = f * cv + (1.f - f) * z;
+    st(cr + (long)t * H, cv);
+    st(hr + (long)t * H, o * cv);
+  }
+}
+
+template <typename T>
+__global__ void qrnn_fo_bwd_kernel(const T* __restrict__ gates,  // activated
+                                   const T* __restrict__ c,
+                                   const T* __restrict__ c0,
+                                   const T* __restrict__ dh,
+                                   const T* __restrict__ dcT,    // carry-in grad
+                                   T* __restrict__ dgates,       // preact grads
+                                   T* __restrict__ dc0,
+                                   int B, int Tn, int H) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * H) return;
+  const int b = idx / H, j = idx % H;
+  const T* g = gates + (long)b * Tn * 3 * H + j;
+  T* dg = dgates + (long)b * Tn * 3 * H + j;
+  const T* cr = c + (long)b * Tn * H + j;
+  const T* dhr = dh + (long)b * Tn * H + j;
+  float dc = ld(dcT + (long)b * H + j);
+  for (int t = Tn - 1; t >= 0; --t) {
+    const T* gt = g + (long)t * 3 * H;
+    const float z = ld(gt), f = ld(gt + H), o = ld(gt + 2 * H);
+    const float ct = ld(cr + (long)t * H);
+    const float cprev = t ? ld(cr + (long)(t - 1) * H)
+                          : ld(c0 + (long)b * H + j);
+    const float dht = ld(dhr + (long)t * H);
+    const float dov = dht * ct;
+    dc += dht * o;
+    const float dzv = dc * (1.f - f);
+    const float dfv = dc * (cprev - z);
+    dc *= f;
+    T* dgt = dg + (long)t * 3 * H;
+    st(dgt, dzv * (1.f - z * z));
+    st(dgt + H, dfv * f * (1.f - f));
+    st(dgt + 2 * H, dov * o * (1.f - o));
+  }
+  st(dc0 + (long)b * H + j, dc);
+}
+
+std::vector<at::Tensor> qrnn_fo_pool_fwd(at::Tensor gates, at::Tensor c0) {
+  CI_CHECK_CUDA(gates); CI_CHECK_CONTIG(gates); CI_CHECK_CONTIG(c0);
+  const int B = gates.size(0), Tn = gates.size(1), H3 = gates.size(2);
+  TORCH_CHECK(H3 % 3 == 0, "gates last dim must be 3*H");
+  const int H = H3 / 3;
+  TORCH_CHECK(c0.size(0) == B && c0.size(1) == H, "c0 shape mismatch");
+  TORCH_CHECK(c0.scalar_type() == gates.scalar_type());
+  auto h = at::empty({B, Tn, H}, gates.options());
+  auto c = at::empty({B, Tn, H}, gates.options());
+  const int threads = 256;
+  const int blocks = ceil_div((long)B * H, threads);
+  CI_DISPATCH_FB(gates.scalar_type(), "qrnn_fo_fwd", [&] {
+    hipLaunchKernelGGL((qrnn_fo_fwd_kernel<scalar_t>), dim3(blocks),
+        dim3(threads), 0, stream(),
+        reinterpret_cast<scalar_t*>(gates.data_ptr()),
+        reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+        reinterpret_cast<scalar_t*>(h.data_ptr()),
+        reinterpret_cast<scalar_t*>(c.data_ptr()), B, Tn, H);
+  });
+  return {h, c};
+}
+
+std::vector<at::Tensor> qrnn_fo_pool_bwd(at::Tensor gates, at::Tensor c,
+                                         at::Tensor c0, at::Tensor dh,
+                                         at::Tensor dcT) {
+  CI_CHECK_CUDA(gates); CI_CHECK_CONTIG(gates); CI_CHECK_CONTIG(c);
+  CI_CHECK_CONTIG(c0); CI_CHECK_CONTIG(dh); CI_CHECK_CONTIG(dcT);
+  const int B = gates.size(0), Tn = gates.size(1), H = gates.size(2) / 3;
+  auto dgates = at::empty_like(gates);
+  auto dc0 = at::empty_like(c0);
+  const int threads = 256;
+  const int blocks = ceil_div((long)B * H, threads);
+  CI_DISPATCH_FB(gates.scalar_type(), "qrnn_fo_bwd", [&] {
+    hipLaunchKernelGGL((qrnn_fo_bwd_kernel<scalar_t>), dim3(blocks),
+        dim3(threads), 0, stream(),
+        reinterpret_cast<const scalar_t*>(gates.data_ptr()),
+        reinterpret_cast<const scalar_t*>(c.data_ptr()),
+        reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+        reinterpret_cast<const scalar_t*>(dh.data_ptr()),
+        reinterpret_cast<const scalar_t*>(dcT.data_ptr()),
+        reinterpret_cast<scalar_t*>(dgates.data_ptr()),
+        reinterpret_cast<scalar_t*>(dc0.data_ptr()), B, Tn, H);
+  });
+  return {dgates, dc0};
+}
+
+}  // namespace ci

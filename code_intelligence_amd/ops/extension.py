@@ -32,6 +32,7 @@ SOURCES = [
     "lstm_gemm.hip",
     "lstm_gemv.hip",
     "pool.hip",
+    "qrnn_pool.hip",
     "adam.hip",
     "ce.hip",
     "embedding.hip",

@@ -34,6 +34,8 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--vocab_sz", type=int, default=60000)
     p.add_argument("--bs", type=int, default=104)
     p.add_argument("--bptt", type=int, default=67)
+    p.add_argument("--qrnn", type=lambda s: s.lower() == "true", default=False,
+                   help="QRNN encoder instead of LSTM (reference train.py:43)")
     p.add_argument("--lr", type=float, default=0.0013)
     p.add_argument("--wd", type=float, default=0.012)
     p.add_argument("--one_cycle", type=lambda s: s.lower() != "false", default=True)
@@ -65,7 +67,7 @@ def main(argv=None) -> dict:
     n_valid = max(1, len(docs) // 10)
     train_docs, valid_docs = docs[n_valid:], docs[:n_valid]
     model = AWDLSTM(vocab_sz=vocab_sz, emb_sz=args.emb_sz, n_hid=args.n_hid,
-                    n_layers=args.n_layers)
+                    n_layers=args.n_layers, qrnn=args.qrnn)
     dtype = torch.bfloat16 if (args.dtype == "bf16" and device != "cpu") else torch.float32
     model = model.to(device=device, dtype=dtype)
     dev = torch.device(device)

@@ -216,3 +216,45 @@ def test_ce_fp8_optin_close_to_bf16():
     finally:
         os.environ.pop("CI_CE_FP8", None)
     assert abs(fp8 - base) / base < 0.01, (base, fp8)
+
+
+def test_qrnn_fo_pool_matches_cpu_fp32():
+    """HIP fo-pool scan fwd+bwd vs the plain-torch reference (fp32)."""
+    from code_intelligence_amd.ops.qrnn import _fo_pool_torch, fo_pool
+    torch.manual_seed(0)
+    B, T, H = 5, 13, 96
+    gates = torch.randn(B, T, 3 * H)
+    c0 = torch.randn(B, H)
+    gc, cc = gates.clone().requires_grad_(), c0.clone().requires_grad_()
+    h_ref, cT_ref = _fo_pool_torch(gc, cc)
+    dh = torch.randn_like(h_ref)
+    dcT = torch.randn_like(cT_ref)
+    (h_ref * dh).sum().add_((cT_ref * dcT).sum()).backward()
+
+    gg = gates.to(DEV).requires_grad_()
+    cg = c0.to(DEV).requires_grad_()
+    h, cT = fo_pool(gg, cg)
+    assert torch.allclose(h.cpu(), h_ref.detach(), atol=1e-5)
+    assert torch.allclose(cT.cpu(), cT_ref.detach(), atol=1e-5)
+    (h * dh.to(DEV)).sum().add_((cT * dcT.to(DEV)).sum()).backward()
+    assert torch.allclose(gg.grad.cpu(), gc.grad, atol=1e-5)
+    assert torch.allclose(cg.grad.cpu(), cc.grad, atol=1e-5)
+
+
+def test_qrnn_layer_gpu_matches_cpu():
+    """Whole QRNN layer (GEMM + scan + window-2 shift) GPU bf16 vs CPU fp32."""
+    from code_intelligence_amd.models import WeightDroppedQRNN
+    torch.manual_seed(1)
+    B, T, E, H = 4, 10, 64, 80
+    layer = WeightDroppedQRNN(E, H, weight_p=0.0, window=2).eval()
+    x = torch.randn(B, T, E)
+    c0 = torch.zeros(B, H)
+    ref, (_, cT_ref) = layer(x, (c0, c0))
+    layer.reset()
+    lg = WeightDroppedQRNN(E, H, weight_p=0.0, window=2).to(DEV, torch.bfloat16).eval()
+    lg.load_state_dict({k: v.to(DEV, torch.bfloat16)
+                        for k, v in layer.state_dict().items()})
+    out, (_, cT) = lg(x.to(DEV, torch.bfloat16),
+                      (c0.to(DEV, torch.bfloat16),) * 2)
+    assert (out.float().cpu() - ref).abs().max() < 0.05
+    assert (cT.float().cpu() - cT_ref).abs().max() < 0.05
