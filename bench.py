@@ -49,6 +49,8 @@ def parse_args():
     p.add_argument("--hid", type=int, default=2400)
     p.add_argument("--layers", type=int, default=4)
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--qrnn", type=lambda v: v.lower() == "true", default=False,
+                   help="bench the QRNN encoder variant (not the headline config)")
     p.add_argument("--lstm_mode", type=str, default=None,
                    help="override CI_LSTM_MODE (fused|lib)")
     return p.parse_args()
@@ -71,7 +73,8 @@ def main():
 
     torch.manual_seed(1234 + rank)
     model = AWDLSTM(vocab_sz=args.vocab, emb_sz=args.emb, n_hid=args.hid,
-                    n_layers=args.layers).to(device=device, dtype=dtype)
+                    n_layers=args.layers, qrnn=args.qrnn
+                    ).to(device=device, dtype=dtype)
     trainer = LMTrainer(model, TrainConfig(), distributed=(world > 1))
     model.train()
     model.reset(args.bs)
@@ -129,7 +132,7 @@ def main():
             "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"AWD-LSTM {args.layers}x{args.hid} emb{args.emb} vocab{args.vocab}",
+                "model": f"AWD-{'QRNN' if args.qrnn else 'LSTM'} {args.layers}x{args.hid} emb{args.emb} vocab{args.vocab}",
                 "global_batch": args.bs * world,
                 "seq_len": args.seq,
                 "parallelism": f"dp{world}",
