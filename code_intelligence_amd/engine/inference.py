@@ -49,7 +49,8 @@ class InferenceWrapper:
             cfg = json.loads((root / "config.json").read_text())
             self.vocab = Vocab.load(root / "vocab.json")
             model = AWDLSTM(vocab_sz=len(self.vocab), emb_sz=cfg["emb_sz"],
-                            n_hid=cfg["n_hid"], n_layers=cfg["n_layers"])
+                            n_hid=cfg["n_hid"], n_layers=cfg["n_layers"],
+                            qrnn=cfg.get("qrnn", False))
             enc_file = root / cfg.get("encoder_file", "encoder.pth")
             model.load_encoder(enc_file)
             self.encoder = model.encoder
@@ -162,6 +163,7 @@ def save_artifacts(model: AWDLSTM, vocab: Vocab, path) -> None:
     enc = model.encoder
     (root / "config.json").write_text(json.dumps({
         "emb_sz": enc.emb_sz, "n_hid": enc.n_hid, "n_layers": enc.n_layers,
+        "qrnn": bool(getattr(enc, "qrnn", False)),
         "encoder_file": "encoder.pth"}))
     vocab.save(root / "vocab.json")
     model.save_encoder(root / "encoder.pth")

@@ -21,9 +21,9 @@ def _make_batches(seed, vocab=64):
     return x, y
 
 
-def _build_model():
+def _build_model(qrnn=False):
     torch.manual_seed(0)
-    m = AWDLSTM(vocab_sz=64, emb_sz=16, n_hid=24, n_layers=2)
+    m = AWDLSTM(vocab_sz=64, emb_sz=16, n_hid=24, n_layers=2, qrnn=qrnn)
     # deterministic: disable dropout noise
     m.eval()
     for p in m.parameters():
@@ -31,13 +31,13 @@ def _build_model():
     return m
 
 
-def _worker(rank, out_path):
+def _worker(rank, out_path, qrnn=False):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(WORLD),
                       MASTER_ADDR="127.0.0.1", MASTER_PORT="29511",
                       LOCAL_RANK=str(rank))
     dist.init_process_group("gloo", rank=rank, world_size=WORLD)
     try:
-        m = _build_model()
+        m = _build_model(qrnn)
         tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
         x, y = _make_batches(seed=100 + rank)
         tr.dist.prepare()
@@ -53,10 +53,12 @@ def _worker(rank, out_path):
 
 
 @pytest.mark.timeout(120)
-def test_ddp_grads_match_single_process_average(tmp_path):
+@pytest.mark.parametrize("qrnn", [False, True], ids=["lstm", "qrnn"])
+def test_ddp_grads_match_single_process_average(tmp_path, qrnn):
     ctx = mp.get_context("spawn")
     out = str(tmp_path / "grads.pt")
-    procs = [ctx.Process(target=_worker, args=(r, out)) for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker, args=(r, out, qrnn))
+             for r in range(WORLD)]
     for p in procs:
         p.start()
     for p in procs:
@@ -65,7 +67,7 @@ def test_ddp_grads_match_single_process_average(tmp_path):
     dist_grads = torch.load(out, weights_only=True)
 
     # single-process reference: average of the two ranks' grads
-    m = _build_model()
+    m = _build_model(qrnn)
     tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
     acc = {}
     for rank in range(WORLD):

@@ -108,3 +108,22 @@ def test_serve_model_uri_resolution(tmp_path, monkeypatch):
     path = _resolve_model_path("gs://models/lm")
     import json as _json
     assert _json.loads(open(f"{path}/config.json").read())["emb_sz"] == 8
+
+
+def test_wrapper_qrnn_encoder_roundtrip(tmp_path):
+    """--qrnn models serve through the same artifact layout (config.json
+    carries the flag; encoder.pth holds the QRNN state dict)."""
+    torch.manual_seed(1)
+    words = [f"w{i}" for i in range(100)]
+    vocab = Vocab(defaults_specials + words)
+    model = AWDLSTM(vocab_sz=len(vocab), emb_sz=16, n_hid=24, n_layers=2,
+                    qrnn=True)
+    save_artifacts(model, vocab, tmp_path / "art")
+    w = InferenceWrapper(model_path=str(tmp_path / "art"), device="cpu")
+    assert w.encoder.qrnn
+    emb = w.get_pooled_features("w1 w2 w3")
+    assert emb.shape == (1, 48) and torch.isfinite(emb).all()
+    # embedding matches the in-memory model's own encoder output
+    model.eval()
+    model.reset(1)
+    assert torch.allclose(emb, w.get_pooled_features("w1 w2 w3"), atol=1e-6)
