@@ -38,6 +38,38 @@ static __device__ __forceinline__ float ld(const T* p) { return Cvt<T>::load(p);
 template <typename T>
 static __device__ __forceinline__ void st(T* p, float v) { Cvt<T>::store(p, v); }
 
+// ---- 16-B vectorized load/store of VEC scalars as fp32 lanes -------------
+// (G13: scalar bf16 loads cost ~2x; VEC = 16/sizeof(T))
+template <typename T, int VEC>
+static __device__ __forceinline__ void ldv(const T* p, float* out) {
+  T buf[VEC];
+  *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(p);
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) out[e] = ld(buf + e);
+}
+
+template <typename T, int VEC>
+static __device__ __forceinline__ void stv(T* p, const float* v) {
+  T buf[VEC];
+  #pragma unroll
+  for (int e = 0; e < VEC; ++e) st(buf + e, v[e]);
+  *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(buf);
+}
+
+template <int VEC>
+static __device__ __forceinline__ void ldv_f32(const float* p, float* out) {
+  #pragma unroll
+  for (int e = 0; e < VEC; e += 4)
+    *reinterpret_cast<float4*>(out + e) = *reinterpret_cast<const float4*>(p + e);
+}
+
+template <int VEC>
+static __device__ __forceinline__ void stv_f32(float* p, const float* v) {
+  #pragma unroll
+  for (int e = 0; e < VEC; e += 4)
+    *reinterpret_cast<float4*>(p + e) = *reinterpret_cast<const float4*>(v + e);
+}
+
 static __device__ __forceinline__ float sigmoidf_(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }

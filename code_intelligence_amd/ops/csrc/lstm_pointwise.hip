@@ -7,38 +7,8 @@
 
 namespace ci {
 
-// VEC consecutive j per thread with 16-B vector loads/stores (G13: scalar
-// bf16 loads cost ~2x; measured 9.9 us vs the ~5.7 us bandwidth floor).
-// Scalar tail path covers H % VEC != 0.
-template <typename T, int VEC>
-static __device__ __forceinline__ void ldv(const T* p, float* out) {
-  T buf[VEC];
-  *reinterpret_cast<int4*>(buf) = *reinterpret_cast<const int4*>(p);
-  #pragma unroll
-  for (int e = 0; e < VEC; ++e) out[e] = ld(buf + e);
-}
-
-template <typename T, int VEC>
-static __device__ __forceinline__ void stv(T* p, const float* v) {
-  T buf[VEC];
-  #pragma unroll
-  for (int e = 0; e < VEC; ++e) st(buf + e, v[e]);
-  *reinterpret_cast<int4*>(p) = *reinterpret_cast<const int4*>(buf);
-}
-
-template <int VEC>
-static __device__ __forceinline__ void ldv_f32(const float* p, float* out) {
-  #pragma unroll
-  for (int e = 0; e < VEC; e += 4)
-    *reinterpret_cast<float4*>(out + e) = *reinterpret_cast<const float4*>(p + e);
-}
-
-template <int VEC>
-static __device__ __forceinline__ void stv_f32(float* p, const float* v) {
-  #pragma unroll
-  for (int e = 0; e < VEC; e += 4)
-    *reinterpret_cast<float4*>(p + e) = *reinterpret_cast<const float4*>(v + e);
-}
+// VEC consecutive j per thread with 16-B vector loads/stores: ldv/stv
+// helpers live in common.h. Scalar tail path covers H % VEC != 0.
 
 // one thread per (b, j-block): gates = xp + rec (+bias); c' = f*c + i*g;
 // h' = o*tanh(c')

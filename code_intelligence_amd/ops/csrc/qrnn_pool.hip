@@ -82,6 +82,89 @@ __global__ void qrnn_fo_bwd_kernel(const T* __restrict__ gates,  // activated
   st(dc0 + (long)b * H + j, dc);
 }
 
+// Vectorized variants: one thread owns VEC consecutive j (16-B packets), so
+// every access in the t-loop is a full-width vector load/store — measured
+// scalar version ran at ~4.2 TB/s; packets close the gap to the HBM floor.
+// Used when H % VEC == 0 (deployed shapes 2400/800 qualify for bf16 and fp32).
+template <typename T>
+__global__ void qrnn_fo_fwd_kernel_vec(T* __restrict__ gates,
+                                       const T* __restrict__ c0,
+                                       T* __restrict__ h,
+                                       T* __restrict__ c,
+                                       int B, int Tn, int H) {
+  constexpr int VEC = 16 / sizeof(T);
+  const int Hv = H / VEC;
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * Hv) return;
+  const int b = idx / Hv, j = (idx % Hv) * VEC;
+  T* g = gates + (long)b * Tn * 3 * H + j;
+  T* hr = h + (long)b * Tn * H + j;
+  T* cr = c + (long)b * Tn * H + j;
+  float cv[VEC], z[VEC], f[VEC], o[VEC];
+  ldv<T, VEC>(c0 + (long)b * H + j, cv);
+  for (int t = 0; t < Tn; ++t) {
+    T* gt = g + (long)t * 3 * H;
+    ldv<T, VEC>(gt, z); ldv<T, VEC>(gt + H, f); ldv<T, VEC>(gt + 2 * H, o);
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      z[e] = tanhf(z[e]); f[e] = sigmoidf_(f[e]); o[e] = sigmoidf_(o[e]);
+      cv[e] = f[e] * cv[e] + (1.f - f[e]) * z[e];
+    }
+    stv<T, VEC>(gt, z); stv<T, VEC>(gt + H, f); stv<T, VEC>(gt + 2 * H, o);
+    stv<T, VEC>(cr + (long)t * H, cv);
+    float hv[VEC];
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) hv[e] = o[e] * cv[e];
+    stv<T, VEC>(hr + (long)t * H, hv);
+  }
+}
+
+template <typename T>
+__global__ void qrnn_fo_bwd_kernel_vec(const T* __restrict__ gates,
+                                       const T* __restrict__ c,
+                                       const T* __restrict__ c0,
+                                       const T* __restrict__ dh,
+                                       const T* __restrict__ dcT,
+                                       T* __restrict__ dgates,
+                                       T* __restrict__ dc0,
+                                       int B, int Tn, int H) {
+  constexpr int VEC = 16 / sizeof(T);
+  const int Hv = H / VEC;
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= (long)B * Hv) return;
+  const int b = idx / Hv, j = (idx % Hv) * VEC;
+  const T* g = gates + (long)b * Tn * 3 * H + j;
+  T* dg = dgates + (long)b * Tn * 3 * H + j;
+  const T* cr = c + (long)b * Tn * H + j;
+  const T* dhr = dh + (long)b * Tn * H + j;
+  float dc[VEC], z[VEC], f[VEC], o[VEC], ct[VEC], cprev[VEC], dht[VEC];
+  ldv<T, VEC>(dcT + (long)b * H + j, dc);
+  for (int t = Tn - 1; t >= 0; --t) {
+    const T* gt = g + (long)t * 3 * H;
+    ldv<T, VEC>(gt, z); ldv<T, VEC>(gt + H, f); ldv<T, VEC>(gt + 2 * H, o);
+    ldv<T, VEC>(cr + (long)t * H, ct);
+    if (t) ldv<T, VEC>(cr + (long)(t - 1) * H, cprev);
+    else   ldv<T, VEC>(c0 + (long)b * H + j, cprev);
+    ldv<T, VEC>(dhr + (long)t * H, dht);
+    float dz[VEC], df[VEC], dov[VEC];
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      dov[e] = dht[e] * ct[e];
+      dc[e] += dht[e] * o[e];
+      dz[e] = dc[e] * (1.f - f[e]);
+      df[e] = dc[e] * (cprev[e] - z[e]);
+      dc[e] *= f[e];
+      dz[e] *= (1.f - z[e] * z[e]);
+      df[e] *= f[e] * (1.f - f[e]);
+      dov[e] *= o[e] * (1.f - o[e]);
+    }
+    T* dgt = dg + (long)t * 3 * H;
+    stv<T, VEC>(dgt, dz); stv<T, VEC>(dgt + H, df);
+    stv<T, VEC>(dgt + 2 * H, dov);
+  }
+  stv<T, VEC>(dc0 + (long)b * H + j, dc);
+}
+
 std::vector<at::Tensor> qrnn_fo_pool_fwd(at::Tensor gates, at::Tensor c0) {
   CI_CHECK_CUDA(gates); CI_CHECK_CONTIG(gates); CI_CHECK_CONTIG(c0);
   const int B = gates.size(0), Tn = gates.size(1), H3 = gates.size(2);
@@ -92,14 +175,25 @@ std::vector<at::Tensor> qrnn_fo_pool_fwd(at::Tensor gates, at::Tensor c0) {
   auto h = at::empty({B, Tn, H}, gates.options());
   auto c = at::empty({B, Tn, H}, gates.options());
   const int threads = 256;
-  const int blocks = ceil_div((long)B * H, threads);
   CI_DISPATCH_FB(gates.scalar_type(), "qrnn_fo_fwd", [&] {
-    hipLaunchKernelGGL((qrnn_fo_fwd_kernel<scalar_t>), dim3(blocks),
-        dim3(threads), 0, stream(),
-        reinterpret_cast<scalar_t*>(gates.data_ptr()),
-        reinterpret_cast<const scalar_t*>(c0.data_ptr()),
-        reinterpret_cast<scalar_t*>(h.data_ptr()),
-        reinterpret_cast<scalar_t*>(c.data_ptr()), B, Tn, H);
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    if (H % VEC == 0) {
+      const int blocks = ceil_div((long)B * (H / VEC), threads);
+      hipLaunchKernelGGL((qrnn_fo_fwd_kernel_vec<scalar_t>), dim3(blocks),
+          dim3(threads), 0, stream(),
+          reinterpret_cast<scalar_t*>(gates.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+          reinterpret_cast<scalar_t*>(h.data_ptr()),
+          reinterpret_cast<scalar_t*>(c.data_ptr()), B, Tn, H);
+    } else {
+      const int blocks = ceil_div((long)B * H, threads);
+      hipLaunchKernelGGL((qrnn_fo_fwd_kernel<scalar_t>), dim3(blocks),
+          dim3(threads), 0, stream(),
+          reinterpret_cast<scalar_t*>(gates.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+          reinterpret_cast<scalar_t*>(h.data_ptr()),
+          reinterpret_cast<scalar_t*>(c.data_ptr()), B, Tn, H);
+    }
   });
   return {h, c};
 }
@@ -113,17 +207,31 @@ std::vector<at::Tensor> qrnn_fo_pool_bwd(at::Tensor gates, at::Tensor c,
   auto dgates = at::empty_like(gates);
   auto dc0 = at::empty_like(c0);
   const int threads = 256;
-  const int blocks = ceil_div((long)B * H, threads);
   CI_DISPATCH_FB(gates.scalar_type(), "qrnn_fo_bwd", [&] {
-    hipLaunchKernelGGL((qrnn_fo_bwd_kernel<scalar_t>), dim3(blocks),
-        dim3(threads), 0, stream(),
-        reinterpret_cast<const scalar_t*>(gates.data_ptr()),
-        reinterpret_cast<const scalar_t*>(c.data_ptr()),
-        reinterpret_cast<const scalar_t*>(c0.data_ptr()),
-        reinterpret_cast<const scalar_t*>(dh.data_ptr()),
-        reinterpret_cast<const scalar_t*>(dcT.data_ptr()),
-        reinterpret_cast<scalar_t*>(dgates.data_ptr()),
-        reinterpret_cast<scalar_t*>(dc0.data_ptr()), B, Tn, H);
+    constexpr int VEC = 16 / sizeof(scalar_t);
+    if (H % VEC == 0) {
+      const int blocks = ceil_div((long)B * (H / VEC), threads);
+      hipLaunchKernelGGL((qrnn_fo_bwd_kernel_vec<scalar_t>), dim3(blocks),
+          dim3(threads), 0, stream(),
+          reinterpret_cast<const scalar_t*>(gates.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dh.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dcT.data_ptr()),
+          reinterpret_cast<scalar_t*>(dgates.data_ptr()),
+          reinterpret_cast<scalar_t*>(dc0.data_ptr()), B, Tn, H);
+    } else {
+      const int blocks = ceil_div((long)B * H, threads);
+      hipLaunchKernelGGL((qrnn_fo_bwd_kernel<scalar_t>), dim3(blocks),
+          dim3(threads), 0, stream(),
+          reinterpret_cast<const scalar_t*>(gates.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c.data_ptr()),
+          reinterpret_cast<const scalar_t*>(c0.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dh.data_ptr()),
+          reinterpret_cast<const scalar_t*>(dcT.data_ptr()),
+          reinterpret_cast<scalar_t*>(dgates.data_ptr()),
+          reinterpret_cast<scalar_t*>(dc0.data_ptr()), B, Tn, H);
+    }
   });
   return {dgates, dc0};
 }
