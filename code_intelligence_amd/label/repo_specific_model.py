@@ -9,6 +9,7 @@ thresholds (None => never predict)."""
 from __future__ import annotations
 
 import logging
+import os
 from typing import Dict, List, Optional
 
 import numpy as np
@@ -28,11 +29,15 @@ EMBEDDING_DIM = 1600
 class RepoSpecificLabelModel(IssueLabelModel):
     def __init__(self, mlp: MLPWrapper, label_names: List[str],
                  thresholds: Dict[int, Optional[float]],
-                 embedding_api_endpoint: str = DEFAULT_EMBEDDING_ENDPOINT,
+                 embedding_api_endpoint: Optional[str] = None,
                  session=None):
         self.mlp = mlp
         self.label_names = label_names
         self.thresholds = thresholds
+        if embedding_api_endpoint is None:
+            # deploy-time wiring like the reference's k8s service DNS name
+            embedding_api_endpoint = os.environ.get(
+                "ISSUE_EMBEDDING_SERVICE", DEFAULT_EMBEDDING_ENDPOINT)
         self.endpoint = embedding_api_endpoint.rstrip("/")
         if session is None:
             import requests
@@ -41,7 +46,7 @@ class RepoSpecificLabelModel(IssueLabelModel):
 
     @classmethod
     def from_repo(cls, repo_owner: str, repo_name: str,
-                  embedding_api_endpoint: str = DEFAULT_EMBEDDING_ENDPOINT,
+                  embedding_api_endpoint: Optional[str] = None,
                   store: Optional[ObjectStore] = None, session=None
                   ) -> "RepoSpecificLabelModel":
         """reference repo_specific_model.py:32-88."""
