@@ -143,3 +143,15 @@ class JSONRunLogger(Callback):
     def on_epoch_end(self, trainer, epoch, metrics):
         self._log({"event": "epoch", "epoch": epoch, **metrics})
         return False
+
+
+class TerminateOnNaN(Callback):
+    """Abort training on the first non-finite loss instead of burning the
+    rest of the schedule (the reference's fastai loop would run to
+    completion on NaN). Raises FloatingPointError with step context."""
+
+    def on_step_end(self, trainer, step, loss):
+        if loss != loss or loss in (float("inf"), float("-inf")):
+            raise FloatingPointError(
+                f"non-finite training loss {loss} at step {step} "
+                f"(lr={getattr(trainer, 'last_lr', None)})")

@@ -118,3 +118,13 @@ def test_label_cli_publish_and_logs(tmp_path, capsys):
                                 "message": "hello", "repo": "kf"}) + "\n")
     label_cli_main(["logs", "--path", str(logf)])
     assert "hello" in capsys.readouterr().out
+
+
+def test_terminate_on_nan():
+    from code_intelligence_amd.train.callbacks import TerminateOnNaN
+    cb = TerminateOnNaN()
+    cb.on_step_end(None, 5, 1.25)  # finite: no-op
+    with pytest.raises(FloatingPointError):
+        cb.on_step_end(None, 6, float("nan"))
+    with pytest.raises(FloatingPointError):
+        cb.on_step_end(None, 7, float("inf"))
