@@ -1,10 +1,11 @@
 from .trainer import LMTrainer, TrainConfig
 from .callbacks import (Callback, CallbackList, EarlyStopping, SaveModel,
-                        ReduceLROnPlateau, CSVLogger, JSONRunLogger)
+                        ReduceLROnPlateau, CSVLogger, JSONRunLogger,
+                        TerminateOnNaN)
 from .schedules import OneCycle, FlatSchedule
 
 __all__ = [
     "LMTrainer", "TrainConfig", "Callback", "CallbackList", "EarlyStopping",
     "SaveModel", "ReduceLROnPlateau", "CSVLogger", "JSONRunLogger",
-    "OneCycle", "FlatSchedule",
+    "TerminateOnNaN", "OneCycle", "FlatSchedule",
 ]
