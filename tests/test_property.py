@@ -1,0 +1,69 @@
+"""Property-based tests (hypothesis) — randomized invariants on the pure
+components: scan recurrence, pooling, tokenizer parity, URL/spec codecs."""
+import string
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from code_intelligence_amd.gh.util import (build_issue_spec, build_issue_url,
+                                           parse_issue_spec, parse_issue_url)
+from code_intelligence_amd.ops.pool import _cpu_concat_pool
+from code_intelligence_amd.ops.qrnn import _fo_pool_torch
+
+NAME = st.text(alphabet=string.ascii_lowercase + string.digits + "-_.",
+               min_size=1, max_size=20).filter(
+    lambda s: not s.startswith(".") and "/" not in s and "#" not in s)
+
+
+@settings(max_examples=50, deadline=None)
+@given(NAME, NAME, st.integers(min_value=1, max_value=10**8))
+def test_issue_spec_url_roundtrip(owner, repo, num):
+    assert parse_issue_spec(build_issue_spec(owner, repo, num)) == (owner, repo, num)
+    assert parse_issue_url(build_issue_url(owner, repo, num)) == (owner, repo, num)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 4), st.integers(1, 9), st.integers(1, 8),
+       st.integers(0, 2**31 - 1))
+def test_fo_pool_invariants(B, T, H, seed):
+    g = torch.Generator().manual_seed(seed)
+    gates = torch.randn(B, T, 3 * H, generator=g, dtype=torch.float64)
+    c0 = torch.randn(B, H, generator=g, dtype=torch.float64)
+    h, cT = _fo_pool_torch(gates, c0)
+    assert h.shape == (B, T, H) and cT.shape == (B, H)
+    # c_t is a convex combination of c_{t-1} and z_t in (-1,1) =>
+    # elementwise |c_T| <= max(|c0|, 1)
+    bound = torch.maximum(c0.abs(), torch.ones_like(c0))
+    assert (cT.abs() <= bound + 1e-9).all()
+    # output gate in (0,1): |h| <= |c|
+    z = torch.tanh(gates[..., :H])
+    assert (h.abs() <= 1 + c0.abs().amax() + 1e-9).all()
+    # T=1 closed form
+    if T == 1:
+        f = torch.sigmoid(gates[:, 0, H:2 * H])
+        o = torch.sigmoid(gates[:, 0, 2 * H:])
+        c1 = f * c0 + (1 - f) * z[:, 0]
+        assert torch.allclose(h[:, 0], o * c1, atol=1e-12)
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.integers(1, 3), st.integers(1, 7), st.integers(1, 5),
+       st.integers(0, 2**31 - 1))
+def test_concat_pool_matches_manual(B, T, H, seed):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(B, T, H, generator=g)
+    lens = torch.randint(1, T + 1, (B,), generator=g, dtype=torch.int32)
+    out = _cpu_concat_pool(x, lens)
+    for b in range(B):
+        n = int(lens[b])
+        ref = torch.cat([x[b, :n].mean(0), x[b, :n].amax(0), x[b, n - 1]])
+        assert torch.allclose(out[b], ref, atol=1e-6)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.text(alphabet=string.printable, max_size=120))
+def test_tokenizer_native_matches_python_ascii(s):
+    from code_intelligence_amd.text.tokenizer import Tokenizer
+    py = Tokenizer(native=False)
+    nat = Tokenizer(native=True)
+    assert nat.process_text(s) == py.process_text(s)
