@@ -77,3 +77,16 @@ for _ in range(3):
 
 torch.cuda.synchronize()
 print("pmc workload done")
+
+# QRNN fo-pool scan fwd/bwd at the deployed shape (B=512, T=16 slice)
+Tq = 16
+qg = torch.randn(512, Tq, 3 * H, device=dev, dtype=dt)
+qc0 = torch.randn(512, H, device=dev, dtype=dt) * 0.1
+for _ in range(3):
+    qh, qc = lib.qrnn_fo_pool_fwd(qg.clone(), qc0)
+qdh = torch.randn_like(qh)
+for _ in range(3):
+    lib.qrnn_fo_pool_bwd(qg, qc, qc0, qdh, torch.zeros_like(qc0))
+
+torch.cuda.synchronize()
+print("pmc workload done (incl qrnn)")
