@@ -82,5 +82,30 @@ def qrnn_forward(x: Tensor, c0: Tensor, weight: Tensor, bias: Tensor,
         inp = x
     else:
         raise ValueError(f"window must be 1 or 2 (got {window})")
-    gates = torch.addmm(bias, inp.reshape(B * T, -1), weight.t())
+    gates = _gates_gemm(inp.reshape(B * T, -1), weight, bias)
     return fo_pool(gates.view(B, T, -1), c0)
+
+
+def _gates_gemm(flat: Tensor, weight: Tensor, bias: Tensor,
+                max_rows: Optional[int] = None) -> Tensor:
+    """bias + flat @ weight.T, chunked over rows when the output would be
+    huge. Large serve batches (e.g. B=200, T=1600) push the fused gate
+    GEMM's output past 2^31 elements, which memory-faults in the GEMM
+    library (observed on ROCm 7.2 at 320000x7200 out); chunking keeps
+    every call well under while leaving each chunk GEMM-saturating."""
+    n_out = weight.size(0)
+    if max_rows is None:
+        max_rows = max(1, ((1 << 31) - (1 << 27)) // n_out)
+    if flat.size(0) <= max_rows:
+        return torch.addmm(bias, flat, weight.t())
+    wt = weight.t()
+    chunks = range(0, flat.size(0), max_rows)
+    if torch.is_grad_enabled() and (flat.requires_grad
+                                    or weight.requires_grad
+                                    or bias.requires_grad):
+        return torch.cat([torch.addmm(bias, flat[s:s + max_rows], wt)
+                          for s in chunks])
+    gates = flat.new_empty((flat.size(0), n_out))
+    for s in chunks:  # serve path: write into one buffer, no autograd
+        torch.addmm(bias, flat[s:s + max_rows], wt, out=gates[s:s + max_rows])
+    return gates

@@ -234,3 +234,18 @@ def test_fp8_weight_serving_close_to_bf16():
     cos = torch.nn.functional.cosine_similarity(base, fp8).item()
     assert cos > 0.995, cos
     assert (base - fp8).abs().max() < 0.1, (base - fp8).abs().max()
+
+
+def test_qrnn_bulk_serve_shape_no_fault():
+    """Regression: B=200 x T=1600 gate GEMM output (2.3e9 elements) used to
+    memory-fault in the GEMM library; the chunked path must survive it."""
+    from code_intelligence_amd.ops.qrnn import qrnn_forward
+    B, T, E, H = 200, 1600, 2400, 2400
+    x = torch.randn(B, T, E, device=DEV, dtype=torch.bfloat16)
+    w = (torch.randn(3 * H, E, device=DEV) * 0.01).to(torch.bfloat16)
+    b = torch.zeros(3 * H, device=DEV, dtype=torch.bfloat16)
+    c0 = torch.zeros(B, H, device=DEV, dtype=torch.bfloat16)
+    with torch.no_grad():
+        h, cT = qrnn_forward(x, c0, w, b, window=1)
+    torch.cuda.synchronize()
+    assert h.shape == (B, T, H) and torch.isfinite(h.float().sum())

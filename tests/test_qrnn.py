@@ -96,3 +96,28 @@ def test_qrnn_window_validation():
     with pytest.raises(ValueError):
         qrnn_forward(torch.randn(1, 2, 4), torch.zeros(1, 3),
                      torch.randn(9, 12), torch.zeros(9), window=3)
+
+
+def test_gates_gemm_chunking_matches_unchunked():
+    from code_intelligence_amd.ops.qrnn import _gates_gemm
+    torch.manual_seed(5)
+    flat = torch.randn(37, 8)
+    w = torch.randn(12, 8)
+    b = torch.randn(12)
+    ref = torch.addmm(b, flat, w.t())
+    # no-grad path (out= chunks)
+    with torch.no_grad():
+        out = _gates_gemm(flat, w, b, max_rows=10)
+    assert torch.allclose(out, ref, atol=1e-6)
+    # autograd path (cat chunks) — gradients flow
+    fw = flat.clone().requires_grad_()
+    ww = w.clone().requires_grad_()
+    out2 = _gates_gemm(fw, ww, b, max_rows=10)
+    assert torch.allclose(out2, ref, atol=1e-6)
+    out2.sum().backward()
+    assert fw.grad is not None and ww.grad is not None
+    ref2 = torch.addmm(b, fw.detach(), ww.detach().t())
+    gref = torch.autograd.functional.vjp(
+        lambda a, c: torch.addmm(b, a, c.t()).sum(), (fw.detach(), ww.detach()))[1]
+    assert torch.allclose(fw.grad, gref[0], atol=1e-6)
+    assert torch.allclose(ww.grad, gref[1], atol=1e-6)
