@@ -13,6 +13,7 @@ per-timestep launch overhead that dominates at serve batch sizes.
 from __future__ import annotations
 
 import json
+import logging
 from pathlib import Path
 from typing import List, Optional, Sequence
 
@@ -22,6 +23,8 @@ import torch
 from ..models.awd_lstm import AWDLSTM, AWDLSTMEncoder
 from ..ops.pool import concat_pool
 from ..text.tokenizer import Tokenizer, Vocab, process_dict as _process_dict
+
+log = logging.getLogger(__name__)
 
 
 class InferenceWrapper:
@@ -60,6 +63,13 @@ class InferenceWrapper:
         self.pad_idx = self.encoder.pad_token
         self._graphs = {}
         self.use_graphs = use_graphs and self.device.type == "cuda"
+        if self.use_graphs and getattr(self.encoder, "qrnn", False):
+            # hipGraph replay of the QRNN encoder memory-faults on ROCm 7.2
+            # (capture of the chunked gate GEMM); eager QRNN serve is already
+            # launch-light, so the opt-in is ignored rather than risked.
+            log.warning("CI_SERVE_GRAPHS ignored for QRNN encoders "
+                        "(graph capture unsupported; serving eagerly)")
+            self.use_graphs = False
 
     # --- reference-parity helpers -----------------------------------------
     def process_dict(self, data: dict) -> dict:

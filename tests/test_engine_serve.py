@@ -127,3 +127,15 @@ def test_wrapper_qrnn_encoder_roundtrip(tmp_path):
     model.eval()
     model.reset(1)
     assert torch.allclose(emb, w.get_pooled_features("w1 w2 w3"), atol=1e-6)
+
+
+def test_wrapper_qrnn_ignores_graph_optin(tmp_path):
+    torch.manual_seed(2)
+    vocab = Vocab(defaults_specials + [f"w{i}" for i in range(50)])
+    model = AWDLSTM(vocab_sz=len(vocab), emb_sz=16, n_hid=24, n_layers=2,
+                    qrnn=True)
+    save_artifacts(model, vocab, tmp_path / "a")
+    w = InferenceWrapper(model_path=str(tmp_path / "a"), device="cpu",
+                         use_graphs=True)
+    assert not w.use_graphs  # downgraded: capture unsupported for QRNN
+    assert w.get_pooled_features("w1 w2").shape == (1, 48)
