@@ -62,7 +62,11 @@ class _FoPoolFn(torch.autograd.Function):
 
 
 def fo_pool(gates: Tensor, c0: Tensor) -> Tuple[Tensor, Tensor]:
-    """(h (B,T,H), c_T (B,H)) from pre-activation gates (B,T,3H)."""
+    """(h (B,T,H), c_T (B,H)) from pre-activation gates (B,T,3H).
+
+    On GPU ``gates`` is CONSUMED: the kernel overwrites it with the
+    activated gate values (saved for backward). Pass a fresh tensor —
+    ``qrnn_forward`` always does."""
     if gates.is_cuda:  # require() inside raises loudly if the .so is missing
         return _FoPoolFn.apply(gates, c0)
     return _fo_pool_torch(gates, c0)
