@@ -128,3 +128,14 @@ def test_terminate_on_nan():
         cb.on_step_end(None, 6, float("nan"))
     with pytest.raises(FloatingPointError):
         cb.on_step_end(None, 7, float("inf"))
+
+
+def test_train_cli_qrnn_end_to_end(tmp_path):
+    m = train_main(["--data_path", "synthetic:60", "--emb_sz", "16",
+                    "--n_hid", "24", "--n_layers", "2", "--vocab_sz", "200",
+                    "--bs", "4", "--bptt", "16", "--epochs", "1",
+                    "--qrnn", "true",
+                    "--model_path", str(tmp_path), "--dtype", "fp32"])
+    assert "valid_loss" in m and m["valid_loss"] == m["valid_loss"]
+    sd = torch.load(tmp_path / "best_enc.pth", weights_only=True)
+    assert "rnns.0.weight_raw" in sd              # QRNN layout persisted
