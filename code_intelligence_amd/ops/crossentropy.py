@@ -107,6 +107,7 @@ class _FusedCEFunction(torch.autograd.Function):
         ctx.save_for_backward(h, weight, b32, tgt64, lse)
         ctx.logits_full = logits_full
         ctx.has_bias = bias is not None
+        ctx.bias_dtype = bias.dtype if bias is not None else None
         return loss
 
     @staticmethod
@@ -141,7 +142,7 @@ class _FusedCEFunction(torch.autograd.Function):
             if has_bias:
                 db += dlog.sum(dim=0).to(torch.float32)
         return (dh, dw.to(weight.dtype),
-                db.to(weight.dtype) if has_bias else None, None)
+                db.to(ctx.bias_dtype) if has_bias else None, None)
 
 
 @torch.no_grad()
