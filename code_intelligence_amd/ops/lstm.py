@@ -190,7 +190,9 @@ class _FusedLSTMFunction(torch.autograd.Function):
                 db_s = dg2.sum(dim=0)
                 _accum_grad(pw, dw_ih_s)
                 _accum_grad(pbi, db_s)
-                _accum_grad(pbh, db_s)
+                # clone: assigning the SAME tensor as both .grads would alias
+                # them and double-count on later accumulation micro-steps
+                _accum_grad(pbh, db_s.clone())
             return (dx_tm.transpose(0, 1), dh0.to(dt), dc0.to(dt),
                     None, dw_hh.to(w_hh.dtype), None, None)
         dw_ih = torch.mm(dg2.t(), x_tm.view(T * B, In))
