@@ -53,6 +53,7 @@ class LMTrainer:
         else:
             self.dist = None
         self.global_step = 0
+        self.epoch = 0  # epochs completed (checkpoint resume)
 
     def _param_groups(self):
         seen, uniq = set(), []
@@ -149,12 +150,17 @@ class LMTrainer:
     # --- checkpoint/resume (full training state; the reference only
     # checkpoints model weights via fastai SaveModelCallback) -------------
     def save_checkpoint(self, path) -> None:
-        torch.save({
+        state = {
             "model": self.model.state_dict(),
             "optimizer": self.opt.state_dict(),
             "lr_scale": self.lr_scale,
             "global_step": self.global_step,
-        }, path)
+            "epoch": self.epoch,
+            "rng": torch.get_rng_state(),
+        }
+        if torch.cuda.is_available():
+            state["rng_cuda"] = torch.cuda.get_rng_state_all()
+        torch.save(state, path)
 
     def load_checkpoint(self, path, map_location="cpu") -> None:
         ckpt = torch.load(path, map_location=map_location, weights_only=False)
@@ -162,6 +168,11 @@ class LMTrainer:
         self.opt.load_state_dict(ckpt["optimizer"])
         self.lr_scale = ckpt.get("lr_scale", 1.0)
         self.global_step = ckpt.get("global_step", 0)
+        self.epoch = ckpt.get("epoch", 0)
+        if "rng" in ckpt:
+            torch.set_rng_state(ckpt["rng"])
+        if "rng_cuda" in ckpt and torch.cuda.is_available():
+            torch.cuda.set_rng_state_all(ckpt["rng_cuda"])
 
     def fit(self, train_loader, valid_loader=None, epochs: int = 1,
             one_cycle: Optional[bool] = None) -> dict:
@@ -177,10 +188,15 @@ class LMTrainer:
         self.cbs.on_train_begin(self)
         metrics: dict = {}
         self.model.train()
-        self.model.reset()
-        step = 0
         t0 = time.time()
-        for epoch in range(epochs):
+        start_epoch = self.epoch  # resume skips completed epochs
+        step = start_epoch * (n_total // epochs) if n_total else 0
+        if start_epoch and hasattr(train_loader, "epoch"):
+            train_loader.epoch = start_epoch  # reproduce the shuffle order
+        for epoch in range(start_epoch, epochs):
+            # fresh hidden state per epoch (fastai RNNTrainer.on_epoch_begin
+            # parity; also makes epoch-granular checkpoint resume exact)
+            self.model.reset()
             losses = []
             for x, y in train_loader:
                 frac = (step / n_total) if n_total else 0.5
@@ -189,6 +205,7 @@ class LMTrainer:
                 losses.append(loss)
                 self.cbs.on_step_end(self, self.global_step, loss)
                 step += 1
+            self.epoch = epoch + 1
             metrics = {"train_loss": sum(losses) / max(len(losses), 1),
                        "time_s": round(time.time() - t0, 2)}
             if valid_loader is not None:

@@ -145,3 +145,39 @@ def test_trainer_checkpoint_resume(tmp_path):
     after_a2 = [a2.train_step(x, y, 1e-3) for _ in range(3)]
     # two resumes from the same checkpoint follow the same trajectory
     assert after_b == pytest.approx(after_a2, abs=1e-6)
+
+
+def test_exact_resume_matches_uninterrupted(tmp_path):
+    """fit 3 epochs straight == fit 1 epoch -> checkpoint -> reload -> fit
+    to 3 (same shuffle order, same dropout RNG, same optimizer state)."""
+    from code_intelligence_amd.data.lm_loader import LMStreamLoader
+    from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
+
+    def mk():
+        torch.manual_seed(7)
+        m = AWDLSTM(vocab_sz=80, emb_sz=12, n_hid=16, n_layers=2)
+        docs = [[(i * 13 + j) % 80 for j in range(30)] for i in range(20)]
+        ld = LMStreamLoader(docs, bs=4, bptt=8, seed=5)
+        return m, ld
+
+    # flat schedule: one-cycle LR depends on the total horizon, so an
+    # interrupted cycle can only match when the cycle length is unchanged
+    cfg = TrainConfig(alpha=0, beta=0, one_cycle=False)
+    m1, ld1 = mk()
+    tr1 = LMTrainer(m1, cfg)
+    tr1.fit(ld1, epochs=3)
+
+    m2, ld2 = mk()
+    tr2 = LMTrainer(m2, cfg)
+    tr2.fit(ld2, epochs=1)
+    tr2.save_checkpoint(tmp_path / "ck.pt")
+
+    m3, ld3 = mk()
+    tr3 = LMTrainer(m3, cfg)
+    tr3.load_checkpoint(tmp_path / "ck.pt")
+    assert tr3.epoch == 1
+    tr3.fit(ld3, epochs=3)
+
+    sd1, sd3 = m1.state_dict(), m3.state_dict()
+    for k in sd1:
+        assert torch.equal(sd1[k], sd3[k]), k
