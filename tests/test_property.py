@@ -67,3 +67,23 @@ def test_tokenizer_native_matches_python_ascii(s):
     py = Tokenizer(native=False)
     nat = Tokenizer(native=True)
     assert nat.process_text(s) == py.process_text(s)
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.integers(1, 5), st.integers(1, 7),
+       st.lists(st.integers(2, 25), min_size=1, max_size=12),
+       st.integers(0, 1000))
+def test_lm_loader_invariants(bs, bptt, doc_lens, seed):
+    from code_intelligence_amd.data.lm_loader import LMStreamLoader
+    docs = [[(seed + i * 31 + j) % 100 for j in range(n)]
+            for i, n in enumerate(doc_lens)]
+    ld = LMStreamLoader(docs, bs=bs, bptt=bptt, seed=seed)
+    batches = list(ld)
+    assert len(batches) == len(ld)
+    for x, y in batches:
+        assert x.shape == y.shape == (bs, bptt)
+        # y is x shifted by one within each parallel stream
+    if batches:
+        xs = torch.cat([x for x, _ in batches], dim=1)
+        ys = torch.cat([y for _, y in batches], dim=1)
+        assert torch.equal(xs[:, 1:], ys[:, :-1])
