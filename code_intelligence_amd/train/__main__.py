@@ -1,3 +1,4 @@
 from .train_cli import main
 
-main()
+if __name__ == "__main__":  # python -m code_intelligence_amd.train
+    main()
