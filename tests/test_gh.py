@@ -246,3 +246,35 @@ def test_github_app_repo_listing_and_reaction(tmp_path):
     assert repos == [{"full_name": "o/r"}]
     out = app.add_reaction("o", "r", 5, "+1", token="tkn")
     assert out["content"] == "+1"
+
+
+def test_token_generator_refreshes_on_expiry():
+    """GitHubAppTokenGenerator: caches until expiry-skew, then re-mints
+    (reference github_app.py:305-364 refresh semantics)."""
+    import datetime
+    from code_intelligence_amd.gh.github_app import GitHubAppTokenGenerator
+
+    class App:
+        def __init__(self):
+            self.mints = 0
+
+        def get_installation_id(self, owner, repo=None):
+            return 42
+
+        def get_installation_access_token(self, iid):
+            self.mints += 1
+            exp = (datetime.datetime.now(datetime.timezone.utc)
+                   + datetime.timedelta(seconds=3600 if self.mints > 1 else 30))
+            return {"token": f"t{self.mints}",
+                    "expires_at": exp.isoformat().replace("+00:00", "Z")}
+
+    app = App()
+    gen = GitHubAppTokenGenerator(app, "org", skew_s=60)
+    assert gen.token == "t1"
+    # first token expires within the 60 s skew -> next access re-mints
+    assert gen.token == "t2"
+    assert app.mints == 2
+    # fresh hour-long token -> cached
+    assert gen.token == "t2"
+    assert app.mints == 2
+    assert gen.auth_headers() == {"Authorization": "token t2"}
