@@ -100,3 +100,29 @@ def test_full_production_path(tmp_path):
         assert l in {"bug", "feature"}
     if added:
         assert gh.labels and gh.comments
+
+
+@pytest.mark.timeout(180)
+def test_bench_contract_cpu_fallback():
+    """bench.py must emit exactly one driver-contract JSON line on stdout
+    (CPU fallback config) — the fields the round driver parses."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+    root = Path(__file__).resolve().parents[1]
+    r = subprocess.run([sys.executable, str(root / "bench.py"),
+                        "--steps", "1", "--warmup", "0"],
+                       capture_output=True, text=True, timeout=150)
+    assert r.returncode == 0, r.stderr[-800:]
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    out = json.loads(lines[0])
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in out, k
+    assert out["metric"] == "LM tokens/sec (whole node)"
+    assert out["scaling"] == "weak" and out["data"] == "synthetic"
+    assert out["value"] > 0 and out["n_gpus"] == 1
+    assert {"model", "global_batch", "seq_len", "parallelism"} <= set(out["config"])
