@@ -4,6 +4,7 @@ routing -> worker applies labels from a queued event. The whole L1-L5
 production path (SURVEY.md §1 data-flow) with only the GitHub REST calls
 faked."""
 import numpy as np
+import pytest
 import torch
 
 from code_intelligence_amd.engine.inference import InferenceWrapper
