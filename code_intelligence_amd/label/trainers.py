@@ -16,6 +16,7 @@ and the repo_mlp notebook semantics:
 from __future__ import annotations
 
 import io
+import os
 import logging
 import tempfile
 from typing import Dict, List, Optional, Tuple
