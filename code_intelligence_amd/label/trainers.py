@@ -100,6 +100,7 @@ def train_repo_mlp(org: str, repo: str, store: Optional[ObjectStore] = None,
     with tempfile.NamedTemporaryFile(suffix=".dpkl", delete=False) as tmp:
         mlp.save_model(tmp.name)
         store.upload(tmp.name, cfg.model_gcs_uri)
+    os.unlink(tmp.name)
     store.write_bytes(cfg.labels_gcs_uri, yaml.safe_dump({
         "labels": names,
         "probability_thresholds": {int(k): v for k, v in thresholds.items()},
