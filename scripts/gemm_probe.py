@@ -12,7 +12,8 @@ import time
 
 import torch
 
-assert torch.cuda.is_available()
+if not torch.cuda.is_available():
+    raise SystemExit(f"{__file__} is a GPU probe workload - run it on an MI355X box (gpurun)")
 dev = "cuda:0"
 torch.manual_seed(0)
 

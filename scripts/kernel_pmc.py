@@ -19,7 +19,8 @@ import torch
 
 from code_intelligence_amd.ops import extension
 
-assert torch.cuda.is_available()
+if not torch.cuda.is_available():
+    raise SystemExit(f"{__file__} is a GPU probe workload - run it on an MI355X box (gpurun)")
 lib = extension.require()
 dev = "cuda:0"
 torch.manual_seed(0)
