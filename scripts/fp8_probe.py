@@ -1,5 +1,7 @@
 import sys; sys.path.insert(0, str(__import__("pathlib").Path(__file__).resolve().parents[1]))
 import time, torch
+if not torch.cuda.is_available():
+    raise SystemExit(f"{__file__} is a GPU probe workload - run it on an MI355X box (gpurun)")
 dev = "cuda:0"
 torch.manual_seed(0)
 C, E, V = 16384, 800, 60000
