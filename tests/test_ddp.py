@@ -85,13 +85,13 @@ def test_ddp_grads_match_single_process_average(tmp_path, qrnn):
         assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
 
 
-def _worker_accum(rank, out_path):
+def _worker_accum(rank, out_path, qrnn=False):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(WORLD),
                       MASTER_ADDR="127.0.0.1", MASTER_PORT="29517",
                       LOCAL_RANK=str(rank))
     dist.init_process_group("gloo", rank=rank, world_size=WORLD)
     try:
-        m = _build_model()
+        m = _build_model(qrnn)
         tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
         mbs = [_make_batches(seed=200 + rank * 10 + i) for i in range(3)]
         tr.train_step(None, None, lr=0.0, micro_batches=mbs)
@@ -104,12 +104,14 @@ def _worker_accum(rank, out_path):
 
 
 @pytest.mark.timeout(120)
-def test_gradient_accumulation_no_sync(tmp_path):
+@pytest.mark.parametrize("qrnn", [False, True], ids=["lstm", "qrnn"])
+def test_gradient_accumulation_no_sync(tmp_path, qrnn):
     """3 micro-batches per rank, all-reduce only on the last: equals the
     single-process mean over all 6 micro-batches."""
     ctx = mp.get_context("spawn")
     out = str(tmp_path / "g.pt")
-    procs = [ctx.Process(target=_worker_accum, args=(r, out)) for r in range(WORLD)]
+    procs = [ctx.Process(target=_worker_accum, args=(r, out, qrnn))
+             for r in range(WORLD)]
     for p in procs:
         p.start()
     for p in procs:
@@ -117,7 +119,7 @@ def test_gradient_accumulation_no_sync(tmp_path):
         assert p.exitcode == 0
     dist_grads = torch.load(out, weights_only=True)
 
-    m = _build_model()
+    m = _build_model(qrnn)
     tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
     acc = {}
     for rank in range(WORLD):
