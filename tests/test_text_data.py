@@ -143,3 +143,14 @@ def test_native_tokenizer_parity():
     # batch API equals per-doc API
     docs = ["a b C", "x " * 30, "sooo COOL"]
     assert nat.process_all(docs) == [nat.process_text(d) for d in docs]
+
+
+def test_vocab_save_load_roundtrip(tmp_path):
+    v = Vocab(defaults_specials + ["alpha", "beta", "gamma"])
+    v.save(tmp_path / "v.json")
+    v2 = Vocab.load(tmp_path / "v.json")
+    assert v2.itos == v.itos
+    assert v2.numericalize(["alpha", "nope", "gamma"]) == \
+        v.numericalize(["alpha", "nope", "gamma"])
+    # pad stays at index 1 (fastai contract the encoder relies on)
+    assert v2.itos[1] == "xxpad"
