@@ -52,11 +52,13 @@ class RecordingGitHub:
         return []
 
 
-def test_full_production_path(tmp_path):
+@pytest.mark.parametrize("qrnn", [False, True], ids=["lstm", "qrnn"])
+def test_full_production_path(tmp_path, qrnn):
     torch.manual_seed(0)
     words = [f"w{i}" for i in range(300)]
     vocab = Vocab(defaults_specials + words)
-    model = AWDLSTM(vocab_sz=len(vocab), emb_sz=16, n_hid=24, n_layers=2)
+    model = AWDLSTM(vocab_sz=len(vocab), emb_sz=16, n_hid=24, n_layers=2,
+                    qrnn=qrnn)
     wrapper = InferenceWrapper(encoder=model.encoder, vocab=vocab, device="cpu")
 
     # archive with a linearly-separable label structure
