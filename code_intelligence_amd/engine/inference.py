@@ -14,6 +14,7 @@ from __future__ import annotations
 
 import json
 import logging
+import os
 from pathlib import Path
 from typing import List, Optional, Sequence
 
@@ -76,7 +77,13 @@ class InferenceWrapper:
         return _process_dict(data, self.tokenizer)
 
     def numericalize(self, text: str) -> List[int]:
-        return self.vocab.numericalize(self.tokenizer.process_text(text))
+        ids = self.vocab.numericalize(self.tokenizer.process_text(text))
+        # opt-in serve-side length cap (CI_SERVE_MAX_TOKENS=N): a pathological
+        # multi-MB issue body otherwise runs unbounded recurrent timesteps.
+        # Default 0 = unlimited (reference parity: inference.py processes
+        # full documents).
+        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "0"))
+        return ids[:cap] if cap > 0 else ids
 
     @staticmethod
     def _bucket(n: int, buckets=(1, 2, 4, 8, 16, 24, 32, 48, 64, 96, 128,

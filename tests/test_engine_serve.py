@@ -139,3 +139,13 @@ def test_wrapper_qrnn_ignores_graph_optin(tmp_path):
                          use_graphs=True)
     assert not w.use_graphs  # downgraded: capture unsupported for QRNN
     assert w.get_pooled_features("w1 w2").shape == (1, 48)
+
+
+def test_serve_max_tokens_cap(tmp_path, monkeypatch):
+    w = _tiny_wrapper(tmp_path)
+    long_text = " ".join(f"w{i % 50}" for i in range(500))
+    assert len(w.numericalize(long_text)) == 500
+    monkeypatch.setenv("CI_SERVE_MAX_TOKENS", "64")
+    assert len(w.numericalize(long_text)) == 64
+    emb = w.get_pooled_features(long_text)
+    assert emb.shape == (1, 48)
