@@ -134,3 +134,52 @@ def test_gradient_accumulation_no_sync(tmp_path, qrnn):
                     acc[n] = acc.get(n, 0) + p_.grad / WORLD
     for n in set(acc) & set(dist_grads):
         assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
+
+
+def _worker_ws4(rank, out_path):
+    os.environ.update(RANK=str(rank), WORLD_SIZE="4",
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29523",
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    try:
+        m = _build_model()
+        tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
+        x, y = _make_batches(seed=300 + rank)
+        tr.dist.prepare()
+        tr.loss_on_batch(x, y).backward()
+        tr.dist.finalize()
+        if rank == 0:
+            grads = {n: p.grad.clone() for n, p in m.named_parameters()
+                     if p.grad is not None}
+            torch.save(grads, out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_ddp_world4_matches_average(tmp_path):
+    """4-rank bucket averaging == single-process mean of the 4 grads
+    (same bucketer path the driver's 8-GPU scale run uses)."""
+    ctx = mp.get_context("spawn")
+    out = str(tmp_path / "g4.pt")
+    procs = [ctx.Process(target=_worker_ws4, args=(r, out)) for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=150)
+        assert p.exitcode == 0
+    dist_grads = torch.load(out, weights_only=True)
+    m = _build_model()
+    tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
+    acc = {}
+    for rank in range(4):
+        for p_ in m.parameters():
+            p_.grad = None
+        m.reset()
+        x, y = _make_batches(seed=300 + rank)
+        tr.loss_on_batch(x, y).backward()
+        for n, p_ in m.named_parameters():
+            if p_.grad is not None:
+                acc[n] = acc.get(n, 0) + p_.grad / 4
+    for n in set(acc) & set(dist_grads):
+        assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
