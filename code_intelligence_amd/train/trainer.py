@@ -176,6 +176,11 @@ class LMTrainer:
 
     def fit(self, train_loader, valid_loader=None, epochs: int = 1,
             one_cycle: Optional[bool] = None) -> dict:
+        """Train up to ``epochs`` TOTAL epochs. ``self.epoch`` (epochs already
+        completed — restored by load_checkpoint, advanced by previous fit
+        calls) is the starting point, so a resumed or repeated fit continues
+        the same trajectory instead of restarting; set ``trainer.epoch = 0``
+        for a fresh run."""
         cfg = self.cfg
         use_oc = cfg.one_cycle if one_cycle is None else one_cycle
         # reference: fit_one_cycle(cycle_len, max_lr=lr*2) (train.py:109-111)
