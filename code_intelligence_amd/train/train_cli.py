@@ -44,6 +44,12 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--resume", type=str, default=None,
+                   help="checkpoint path (trainer.save_checkpoint) to resume "
+                        "from: restores model+optimizer+epoch+RNG and "
+                        "continues the same trajectory")
+    p.add_argument("--save_checkpoint", type=str, default=None,
+                   help="write a full resume checkpoint here after training")
     return p
 
 
@@ -84,8 +90,12 @@ def main(argv=None) -> dict:
         CSVLogger(out / "history.csv"),
         JSONRunLogger(out / "run.jsonl", config=vars(args)),
     ])
+    if args.resume:
+        trainer.load_checkpoint(args.resume, map_location=device)
     metrics = trainer.fit(train_loader, valid_loader,
                           epochs=args.epochs * max(args.cycle_len, 1))
+    if args.save_checkpoint:
+        trainer.save_checkpoint(args.save_checkpoint)
     print(json.dumps({"final": metrics}))
     return metrics
 

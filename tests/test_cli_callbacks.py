@@ -139,3 +139,14 @@ def test_train_cli_qrnn_end_to_end(tmp_path):
     assert "valid_loss" in m and m["valid_loss"] == m["valid_loss"]
     sd = torch.load(tmp_path / "best_enc.pth", weights_only=True)
     assert "rnns.0.weight_raw" in sd              # QRNN layout persisted
+
+
+def test_train_cli_resume_continues(tmp_path):
+    common = ["--data_path", "synthetic-markov:80", "--emb_sz", "16",
+              "--n_hid", "24", "--n_layers", "2", "--vocab_sz", "150",
+              "--bs", "4", "--bptt", "16", "--dtype", "fp32",
+              "--one_cycle", "false", "--model_path", str(tmp_path)]
+    ck = str(tmp_path / "resume.pt")
+    m1 = train_main(common + ["--epochs", "1", "--save_checkpoint", ck])
+    m2 = train_main(common + ["--epochs", "3", "--resume", ck])
+    assert m2["valid_loss"] < m1["valid_loss"]  # training continued downhill
