@@ -46,6 +46,26 @@ def _find_so() -> Path | None:
     return None
 
 
+def _warn_if_stale(so: Path) -> None:
+    """A silently-stale binary is the worst failure mode of in-tree builds:
+    edited kernels that never run. Warn (once, at load) when any csrc file
+    is newer than the .so."""
+    try:
+        so_m = so.stat().st_mtime
+        newer = [f.name for f in _CSRC.iterdir()
+                 if f.suffix in (".hip", ".cpp", ".h")
+                 and not f.name.endswith("_hip.hip")
+                 and f.stat().st_mtime > so_m]
+        if newer:
+            import warnings
+            warnings.warn(
+                f"_ci_hip.so is older than kernel sources {newer}; "
+                "rebuild with python -m code_intelligence_amd.ops.build",
+                RuntimeWarning, stacklevel=3)
+    except OSError:
+        pass
+
+
 def load(required: bool = False):
     """Import the in-tree .so. required=True -> raise if absent."""
     global _ext, _tried
@@ -55,6 +75,8 @@ def load(required: bool = False):
         return None
     _tried = True
     so = _find_so()
+    if so is not None:
+        _warn_if_stale(so)
     if so is None:
         if required:
             raise RuntimeError(
