@@ -61,6 +61,8 @@ def load_docs(data_path: str, vocab_sz: int):
         return docs, vocab_sz
     root = Path(data_path)
     docs = torch.load(root / "docs.pt", weights_only=True)
+    # docs.pt is either the compact {flat, offsets} corpus (prepare_data)
+    # or a legacy list-of-lists; both feed LMStreamLoader
     vocab = json.loads((root / "vocab.json").read_text())
     return docs, len(vocab)
 
@@ -70,8 +72,13 @@ def main(argv=None) -> dict:
     torch.manual_seed(args.seed)
     device = args.device or ("cuda" if torch.cuda.is_available() else "cpu")
     docs, vocab_sz = load_docs(args.data_path, args.vocab_sz)
-    n_valid = max(1, len(docs) // 10)
-    train_docs, valid_docs = docs[n_valid:], docs[:n_valid]
+    if isinstance(docs, dict):
+        from ..data.lm_loader import split_compact
+        n_docs = docs["offsets"].numel() - 1
+        valid_docs, train_docs = split_compact(docs, max(1, n_docs // 10))
+    else:
+        n_valid = max(1, len(docs) // 10)
+        train_docs, valid_docs = docs[n_valid:], docs[:n_valid]
     model = AWDLSTM(vocab_sz=vocab_sz, emb_sz=args.emb_sz, n_hid=args.n_hid,
                     n_layers=args.n_layers, qrnn=args.qrnn)
     dtype = torch.bfloat16 if (args.dtype == "bf16" and device != "cpu") else torch.float32

@@ -28,6 +28,7 @@ from pathlib import Path
 
 import torch
 
+from code_intelligence_amd.data.lm_loader import docs_to_compact
 from code_intelligence_amd.text.tokenizer import (Tokenizer, Vocab,
                                                   process_dict)
 
@@ -75,7 +76,10 @@ def main():
     docs = [vocab.numericalize(t) for t in token_docs]
     out = Path(args.out)
     out.mkdir(parents=True, exist_ok=True)
-    torch.save(docs, out / "docs.pt")
+    # compact corpus: flat int32 + offsets — 8x less RAM than list-of-lists
+    # at the reference's 16.7M-issue scale; LMStreamLoader consumes it
+    # directly (data/lm_loader.py)
+    torch.save(docs_to_compact(docs), out / "docs.pt")
     vocab.save(out / "vocab.json")
     n_tok = sum(len(d) for d in docs)
     # corpus stats (the counts the reference's 01 notebook reports)

@@ -84,14 +84,19 @@ def test_prepare_data_script_end_to_end(tmp_path):
         capture_output=True, text=True, cwd=".")
     assert r.returncode == 0, r.stderr
     docs = torch.load(out / "docs.pt", weights_only=True)
-    assert len(docs) == 40
+    # compact corpus format: flat int32 + per-doc offsets
+    assert docs["offsets"].numel() - 1 == 40
     from code_intelligence_amd.text.tokenizer import Vocab
     v = Vocab.load(out / "vocab.json")
-    assert all(0 <= t < len(v) for d in docs for t in d)
-    # consumable by the train CLI loader
+    assert int(docs["flat"].min()) >= 0
+    assert int(docs["flat"].max()) < len(v)
+    # consumable by the train CLI loader + stream loader
     from code_intelligence_amd.train.train_cli import load_docs
     docs2, vsz = load_docs(str(out), 0)
-    assert vsz == len(v) and len(docs2) == 40
+    assert vsz == len(v)
+    ld = LMStreamLoader(docs2, bs=2, bptt=8)
+    x, y = next(iter(ld))
+    assert x.shape == (2, 8) and x.dtype == torch.int64
 
 
 def test_tokenizer_fuzz_no_crash():
@@ -154,3 +159,23 @@ def test_vocab_save_load_roundtrip(tmp_path):
         v.numericalize(["alpha", "nope", "gamma"])
     # pad stays at index 1 (fastai contract the encoder relies on)
     assert v2.itos[1] == "xxpad"
+
+
+def test_compact_corpus_equivalent_to_lists():
+    from code_intelligence_amd.data.lm_loader import (docs_to_compact,
+                                                      split_compact)
+    docs = [[(i * 7 + j) % 90 + 9 for j in range(5 + i % 11)]
+            for i in range(40)]
+    compact = docs_to_compact(docs)
+    a = LMStreamLoader(docs, bs=4, bptt=6, seed=3)
+    b = LMStreamLoader(compact, bs=4, bptt=6, seed=3)
+    assert len(a) == len(b)
+    for (xa, ya), (xb, yb) in zip(a, b):
+        assert torch.equal(xa, xb) and torch.equal(ya, yb)
+    # split: first 10 docs / rest, content preserved
+    first, rest = split_compact(compact, 10)
+    assert first["offsets"].numel() - 1 == 10
+    assert rest["offsets"].numel() - 1 == 30
+    re_flat = torch.cat([first["flat"], rest["flat"]])
+    assert torch.equal(re_flat, compact["flat"])
+    assert int(rest["offsets"][0]) == 0
