@@ -135,7 +135,15 @@ class InferenceWrapper:
         return self.texts_to_embedding(texts, bs=bs)
 
     def texts_to_embedding(self, texts: Sequence[str], bs: int = 100) -> np.ndarray:
-        docs = [self.numericalize(t) or [0] for t in texts]
+        # batched tokenization (native GIL-released core underneath)
+        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "0"))
+        tok_lists = self.tokenizer.process_all(list(texts))
+        docs = []
+        for toks in tok_lists:
+            ids = self.vocab.numericalize(toks)
+            if cap > 0:
+                ids = ids[:cap]
+            docs.append(ids or [0])
         order = sorted(range(len(docs)), key=lambda i: len(docs[i]))
         out = np.empty((len(docs), 3 * self.emb_sz), dtype=np.float32)
         i = 0
