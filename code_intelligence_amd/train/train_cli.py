@@ -60,7 +60,12 @@ def load_docs(data_path: str, vocab_sz: int):
                                       markov="markov" in data_path)
         return docs, vocab_sz
     root = Path(data_path)
-    docs = torch.load(root / "docs.pt", weights_only=True)
+    try:
+        # compact corpora memory-map straight from disk (no RAM copy of
+        # the flat token tensor until windows touch it)
+        docs = torch.load(root / "docs.pt", weights_only=True, mmap=True)
+    except (RuntimeError, TypeError):
+        docs = torch.load(root / "docs.pt", weights_only=True)
     # docs.pt is either the compact {flat, offsets} corpus (prepare_data)
     # or a legacy list-of-lists; both feed LMStreamLoader
     vocab = json.loads((root / "vocab.json").read_text())
