@@ -210,7 +210,13 @@ class Vocab:
     @classmethod
     def create(cls, tokens: Iterable[List[str]], max_vocab: int = 60000,
                min_freq: int = 2) -> "Vocab":
-        counts = Counter(t for doc in tokens for t in doc)
+        return cls.from_counter(Counter(t for doc in tokens for t in doc),
+                                max_vocab=max_vocab, min_freq=min_freq)
+
+    @classmethod
+    def from_counter(cls, counts: Counter, max_vocab: int = 60000,
+                     min_freq: int = 2) -> "Vocab":
+        """Build from pre-accumulated counts (streaming prepare_data)."""
         itos = [s for s, c in counts.most_common(max_vocab)
                 if c >= min_freq and s not in defaults_specials]
         itos = defaults_specials + itos

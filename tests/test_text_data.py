@@ -179,3 +179,23 @@ def test_compact_corpus_equivalent_to_lists():
     re_flat = torch.cat([first["flat"], rest["flat"]])
     assert torch.equal(re_flat, compact["flat"])
     assert int(rest["offsets"][0]) == 0
+
+
+def test_prepare_data_streaming_matches_in_memory(tmp_path):
+    """--streaming (two-pass bounded-memory) produces a byte-identical
+    corpus + vocab to the in-memory path."""
+    import subprocess
+    import sys
+    a, b = tmp_path / "mem", tmp_path / "str"
+    for extra, out in ((["--workers", "2"], a),
+                       (["--streaming", "--chunk", "16", "--workers", "2"], b)):
+        r = subprocess.run(
+            [sys.executable, "scripts/prepare_data.py", "--archive",
+             "synthetic:60", "--out", str(out), "--max_vocab", "300"] + extra,
+            capture_output=True, text=True, cwd=".")
+        assert r.returncode == 0, r.stderr
+    da = torch.load(a / "docs.pt", weights_only=True)
+    db = torch.load(b / "docs.pt", weights_only=True)
+    assert torch.equal(da["flat"], db["flat"])
+    assert torch.equal(da["offsets"], db["offsets"])
+    assert (a / "vocab.json").read_text() == (b / "vocab.json").read_text()
