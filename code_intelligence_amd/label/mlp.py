@@ -56,13 +56,22 @@ class MLPWrapper:
     def __init__(self, clf: Optional[MLPHead] = None, in_dim: int = 1600,
                  hidden: Sequence[int] = (600, 600), n_labels: int = 1,
                  max_iter: int = 3000, lr: float = 1e-3, device: str = "cpu",
-                 early_stopping: bool = True):
+                 early_stopping: bool = True,
+                 precision_threshold: float = 0.7,
+                 recall_threshold: float = 0.5):
         self.in_dim, self.hidden, self.n_labels = in_dim, tuple(hidden), n_labels
         self.max_iter, self.lr = max_iter, lr
         self.device = torch.device(device)
         self.clf = clf or MLPHead(in_dim, hidden, n_labels)
         self.early_stopping = early_stopping
+        # per-instance thresholds (reference mlp.py:19-20,37-38 ctor kwargs)
+        self.precision_threshold = precision_threshold
+        self.recall_threshold = recall_threshold
         self.probability_thresholds: Dict[int, Optional[float]] = {}
+        # per-label metrics at the chosen threshold (reference mlp.py:41-43)
+        self.precisions: Dict[int, float] = {}
+        self.recalls: Dict[int, float] = {}
+        self.total_labels_count: Optional[int] = None
 
     # --- training ---------------------------------------------------------
     def fit(self, X: np.ndarray, y: np.ndarray, epochs: Optional[int] = None,
@@ -113,27 +122,45 @@ class MLPWrapper:
     predict_proba = predict_probabilities
 
     # --- threshold search (mlp.py:65-98 semantics) ------------------------
-    def find_probability_thresholds(self, X: np.ndarray, y: np.ndarray
+    def find_probability_thresholds(self, X: np.ndarray, y: np.ndarray,
+                                    test_size: float = 0.3
                                     ) -> Dict[int, Optional[float]]:
+        """Reference semantics (mlp.py:65-98): hold out ``test_size`` of the
+        data (random_state 1234), REFIT on the rest, and search each label's
+        P-R curve on the held-out part; stores ``probability_thresholds``,
+        ``precisions``, ``recalls`` (0.0 when no point qualifies) and
+        ``total_labels_count``. Pass ``test_size=0`` to search on the given
+        data with the already-fitted model instead."""
         from sklearn.metrics import precision_recall_curve
-        probs = self.predict_probabilities(X)
         y = np.asarray(y).copy()  # torch.as_tensor needs writable arrays
         if y.ndim == 1:
             y = y[:, None]
+        X = np.asarray(X)
+        if test_size and len(X) >= 4:
+            from sklearn.model_selection import train_test_split
+            X_tr, X_te, y_tr, y_te = train_test_split(
+                X, y, test_size=test_size, random_state=1234)
+            self.fit(X_tr, y_tr)
+        else:
+            X_te, y_te = X, y
+        probs = self.predict_probabilities(X_te)
         out: Dict[int, Optional[float]] = {}
-        for li in range(y.shape[1]):
-            yt, pp = y[:, li], probs[:, li]
-            if yt.sum() == 0 or yt.sum() == len(yt):
-                out[li] = None
-                continue
-            prec, rec, thr = precision_recall_curve(yt, pp)
-            best_thr, best_prec = None, -1.0
-            # thresholds align with prec[1:], rec[1:]
-            for p_, r_, t_ in zip(prec[1:], rec[1:], thr):
-                if p_ >= self.precision_threshold and r_ >= self.recall_threshold:
-                    if p_ > best_prec:
-                        best_prec, best_thr = p_, float(t_)
+        self.precisions, self.recalls = {}, {}
+        self.total_labels_count = y_te.shape[1]
+        for li in range(y_te.shape[1]):
+            yt, pp = y_te[:, li], probs[:, li]
+            best_p, best_r, best_thr = 0.0, 0.0, None
+            if 0 < yt.sum() < len(yt):
+                prec, rec, thr = precision_recall_curve(yt, pp)
+                # sklearn pairing: precision[i]/recall[i] are the metrics of
+                # predicting score >= thresholds[i] (last P/R point has no
+                # threshold) — reference mlp.py:84 zips [:-1] with thresholds
+                for p_, r_, t_ in zip(prec[:-1], rec[:-1], thr):
+                    if p_ >= self.precision_threshold                             and r_ >= self.recall_threshold and p_ > best_p:
+                        best_p, best_r, best_thr = p_, r_, float(t_)
             out[li] = best_thr  # None => never predict this label
+            self.precisions[li] = float(best_p)
+            self.recalls[li] = float(best_r)
         self.probability_thresholds = out
         return out
 

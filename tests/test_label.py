@@ -53,11 +53,15 @@ def test_threshold_search_excludes_unsatisfiable_labels():
                       [.1, .2, .5], [.2, .1, .5], [.15, .3, .5]],
                      dtype=np.float32)
     w.predict_probabilities = lambda X_: probs  # type: ignore
-    thr = w.find_probability_thresholds(X, y)
+    thr = w.find_probability_thresholds(X, y, test_size=0)
     assert thr[0] is not None        # label 0 perfectly separable
     assert thr[1] is None            # label 1: probs anti-correlated
     # label 2: constant 0.5 prob, precision 0.5 < 0.7 -> None
     assert thr[2] is None
+    # reference test_mlp.py:56-59 also inspects the stored metrics
+    assert w.precisions[0] >= 0.7 and w.recalls[0] >= 0.5
+    assert w.precisions[1] == 0.0 and w.recalls[2] == 0.0
+    assert w.total_labels_count == 3
 
 
 def test_mlp_save_load_roundtrip(tmp_path):
