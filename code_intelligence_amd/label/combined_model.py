@@ -20,7 +20,13 @@ class CombinedLabelModels(IssueLabelModel):
         return self._combine_predictions(preds)
 
     @staticmethod
-    def _combine_predictions(predictions: List[Dict[str, float]]) -> Dict[str, float]:
+    def _combine_predictions(predictions, right: Optional[Dict[str, float]] = None
+                             ) -> Dict[str, float]:
+        """Per-label max merge. Accepts either a list of prediction dicts or
+        the reference's pairwise form ``_combine_predictions(left, right)``
+        (combined_model.py:41-54, exercised by its combined_model_test)."""
+        if right is not None or isinstance(predictions, dict):
+            predictions = [predictions or {}, right or {}]
         out: Dict[str, float] = {}
         for p in predictions:
             for label, prob in (p or {}).items():

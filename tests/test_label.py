@@ -389,3 +389,12 @@ def test_github_issue_client_rest_contract():
     assert calls[0][3]["Authorization"] == "token sekrit"
     assert calls[1][1].endswith("/repos/o/r/issues/5/comments")
     assert calls[2][0] == "GET"
+
+
+def test_combine_predictions_pairwise_reference_form():
+    """reference combined_model_test.py:10-31 calls the static method with
+    (left, right) dicts."""
+    left = {"a": .1, "b": .9, "c": .5}
+    right = {"a": .9, "b": .1, "d": .6}
+    got = CombinedLabelModels._combine_predictions(left, right)
+    assert got == {"a": .9, "b": .9, "c": .5, "d": .6}
