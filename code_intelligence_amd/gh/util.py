@@ -16,11 +16,13 @@ ISSUE_URL_RE = re.compile(
     r"https?://github\.com/([^/]+)/([^/]+)/issues/(\d+)/?")
 
 
-def parse_issue_spec(spec: str) -> Optional[Tuple[str, str, int]]:
-    """'owner/repo#1234' -> (owner, repo, 1234); None if malformed."""
+def parse_issue_spec(spec: str) -> Tuple[Optional[str], Optional[str],
+                                         Optional[int]]:
+    """'owner/repo#1234' -> (owner, repo, 1234); (None, None, None) if
+    malformed — unpacking-safe, reference util_test.py:8-25 semantics."""
     m = ISSUE_SPEC_RE.fullmatch(spec or "")
     if not m:
-        return None
+        return None, None, None
     return m.group(1), m.group(2), int(m.group(3))
 
 

@@ -24,10 +24,9 @@ def cmd_get_issue(args):
 
 
 def cmd_label_issue(args):
-    spec = parse_issue_spec(args.issue)
-    if spec is None:
+    owner, repo, num = parse_issue_spec(args.issue)
+    if owner is None:
         sys.exit(f"bad issue spec: {args.issue}")
-    owner, repo, num = spec
     q = LocalQueue(spool_path=args.spool)
     mid = q.publish(repo_owner=owner, repo_name=repo, issue_num=num,
                     installation_id=args.installation_id or "")

@@ -47,7 +47,9 @@ class FakeSession:
 
 def test_parse_issue_spec_and_url():
     assert parse_issue_spec("kubeflow/tf-operator#123") == ("kubeflow", "tf-operator", 123)
-    assert parse_issue_spec("garbage") is None
+    # malformed -> unpackable (None, None, None): reference util_test.py:8-25
+    assert parse_issue_spec("garbage") == (None, None, None)
+    assert parse_issue_spec("kubeflow/tfjob/tfjob") == (None, None, None)
     assert parse_issue_url("https://github.com/a/b/issues/7") == ("a", "b", 7)
     assert build_issue_url("a", "b", 7) == "https://github.com/a/b/issues/7"
 
