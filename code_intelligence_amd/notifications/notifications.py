@@ -89,6 +89,15 @@ class NotificationManager:
         return marked
 
 
+def process_issue_results(data: dict) -> List[dict]:
+    """Unpack one issues-query page into its issue nodes
+    (reference notifications.py:44-60; its notifications_test asserts the
+    node list shape)."""
+    edges = (((data.get("data") or {}).get("repository") or {})
+             .get("issues") or {}).get("edges") or []
+    return [e["node"] for e in edges]
+
+
 def download_issues(repo: str, output, client: Optional[GraphQLClient] = None,
                     page_size: int = 100) -> List[dict]:
     """Bulk-download issues + first comments into shard files."""

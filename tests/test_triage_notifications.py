@@ -194,3 +194,15 @@ def test_archive_max_age_filter(tmp_path):
     bigquery.write_archive_events(events, tmp_path / "e.jsonl")
     df = bigquery.get_issues("o", max_age_days=30, archive_root=tmp_path)
     assert df["issue_num"].tolist() == [1]
+
+
+def test_process_issue_results():
+    """reference notifications_test.py:6-12 contract."""
+    from code_intelligence_amd.notifications.notifications import (
+        process_issue_results)
+    page = {"data": {"repository": {"issues": {
+        "edges": [{"node": {"title": f"t{i}", "number": i}}
+                  for i in range(100)]}}}}
+    issues = process_issue_results(page)
+    assert len(issues) == 100
+    assert "title" in issues[0]
