@@ -160,9 +160,10 @@ class Worker:
         merged = {**org_cfg, **repo_cfg}
         predictions = self.apply_repo_config(merged, predictions)
 
-        if issue_data is None and self.predictor.client is not None:
+        client = getattr(self.predictor, "client", None)
+        if issue_data is None and client is not None:
             issue_data = github_util.get_issue(
-                build_issue_url(owner, repo, issue_num), self.predictor.client)
+                build_issue_url(owner, repo, issue_num), client)
         issue_data = issue_data or {"labels": [], "removed_labels": []}
         existing = set(issue_data.get("labels") or [])
         removed = set(issue_data.get("removed_labels") or [])
