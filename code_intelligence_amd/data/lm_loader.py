@@ -77,7 +77,7 @@ class LMStreamLoader:
     def _stream(self) -> Tensor:
         n = self.n_docs
         if n == 0:
-            return torch.empty(0, dtype=torch.int64)
+            return torch.empty(0, dtype=torch.int32)
         if self.shuffle:
             g = torch.Generator().manual_seed(self.seed + self.epoch)
             order = torch.randperm(n, generator=g)
@@ -89,7 +89,9 @@ class LMStreamLoader:
         seg = L + bos
         dst_start = torch.zeros(n + 1, dtype=torch.int64)
         torch.cumsum(seg, 0, out=dst_start[1:])
-        out = torch.empty(int(dst_start[-1]), dtype=torch.int64)
+        # int32 epoch stream (windows cast to int64 at yield): halves the
+        # resident stream at reference scale (~9 vs ~19 GB for 2.3B tokens)
+        out = torch.empty(int(dst_start[-1]), dtype=torch.int32)
         src_start = self._offsets[:-1][order]
         # vectorized permuted-concat in doc chunks (bounds index-tensor RAM)
         CH = 262144
@@ -102,7 +104,7 @@ class LMStreamLoader:
                 torch.arange(e - s, dtype=torch.int64), seg_c)
             local = pos - (dst_start[s:e][doc] - base)
             src = src_start[s:e][doc] + (local - bos)
-            vals = self._flat[src.clamp_min_(0)].to(torch.int64)
+            vals = self._flat[src.clamp_min_(0)].to(torch.int32)
             if bos:
                 vals[local == 0] = self.bos_idx
             out[base: base + pos.numel()] = vals
@@ -130,6 +132,9 @@ class LMStreamLoader:
             x = mat[:, s: s + self.bptt]
             y = mat[:, s + 1: s + self.bptt + 1]
             if self.device is not None:
+                # copy the int32 window first (async from the pinned mat),
+                # widen to int64 on-device
                 x = x.to(self.device, non_blocking=True)
                 y = y.to(self.device, non_blocking=True)
-            yield x.contiguous(), y.contiguous()
+            yield x.contiguous().to(torch.int64), \
+                y.contiguous().to(torch.int64)
