@@ -31,9 +31,12 @@ class Message:
     message_id: str = field(default_factory=lambda: uuid.uuid4().hex)
     _acked: bool = False
     _nacked: bool = False
+    _on_ack: Optional[Callable[[str], None]] = None
 
     def ack(self):
         self._acked = True
+        if self._on_ack is not None:
+            self._on_ack(self.message_id)
 
     def nack(self):
         self._nacked = True
@@ -67,6 +70,15 @@ class LocalQueue(BaseQueue):
         self.q: "queue.Queue[Message]" = queue.Queue()
         self.spool = Path(spool_path) if spool_path else None
         self._spool_pos = 0
+        self._ack_lock = threading.Lock()
+        self._acked_ids: set = set()
+        if self.spool is not None:
+            # durable acks: a restarted consumer re-delivers only UNacked
+            # messages (Pub/Sub at-least-once semantics, not replay-all)
+            self._ack_path = self.spool.with_suffix(self.spool.suffix + ".acks")
+            if self._ack_path.exists():
+                self._acked_ids = set(
+                    self._ack_path.read_text().split())
 
     def publish(self, data: bytes = b"", **attributes) -> str:
         msg = Message(data=data, attributes={k: str(v) for k, v in attributes.items()})
@@ -86,10 +98,22 @@ class LocalQueue(BaseQueue):
             lines = f.readlines()
         for line in lines[self._spool_pos:]:
             obj = json.loads(line)
+            if obj["id"] in self._acked_ids:
+                continue  # durably acked before a restart
             self.q.put(Message(data=obj["data"].encode(),
                                attributes=obj["attributes"],
-                               message_id=obj["id"]))
+                               message_id=obj["id"],
+                               _on_ack=self._record_ack))
         self._spool_pos = len(lines)
+
+    def _record_ack(self, message_id: str) -> None:
+        if self.spool is None:
+            return
+        with self._ack_lock:
+            if message_id not in self._acked_ids:
+                self._acked_ids.add(message_id)
+                with open(self._ack_path, "a") as f:
+                    f.write(message_id + "\n")
 
     def pull(self, timeout: Optional[float] = None) -> Optional[Message]:
         self._drain_spool()

@@ -398,3 +398,31 @@ def test_combine_predictions_pairwise_reference_form():
     right = {"a": .9, "b": .1, "d": .6}
     got = CombinedLabelModels._combine_predictions(left, right)
     assert got == {"a": .9, "b": .9, "c": .5, "d": .6}
+
+
+def test_spool_queue_durable_acks(tmp_path):
+    """A restarted consumer re-delivers only UNacked messages (Pub/Sub
+    at-least-once semantics — not replay-all)."""
+    from code_intelligence_amd.label.queueing import LocalQueue
+    spool = tmp_path / "q.jsonl"
+    q1 = LocalQueue(spool_path=spool)
+    for i in range(30):
+        q1.publish(repo_owner="o", repo_name="r", issue_num=i)
+    for _ in range(12):
+        m = q1.pull(timeout=0.05)
+        m.ack()
+    # message 13 pulled but NOT acked -> must come back after restart
+    q1.pull(timeout=0.05)
+    del q1
+    q2 = LocalQueue(spool_path=spool)
+    nums = []
+    while True:
+        m = q2.pull(timeout=0.05)
+        if m is None:
+            break
+        nums.append(int(m.attributes["issue_num"]))
+        m.ack()
+    assert nums == list(range(12, 30))  # unacked 12..29 redelivered once
+    del q2
+    q3 = LocalQueue(spool_path=spool)  # everything acked now
+    assert q3.pull(timeout=0.05) is None
