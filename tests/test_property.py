@@ -87,3 +87,41 @@ def test_lm_loader_invariants(bs, bptt, doc_lens, seed):
         xs = torch.cat([x for x, _ in batches], dim=1)
         ys = torch.cat([y for _, y in batches], dim=1)
         assert torch.equal(xs[:, 1:], ys[:, :-1])
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.booleans(), st.booleans(), st.booleans(), st.booleans(),
+       st.booleans(), st.booleans())
+def test_triage_rules_property(has_kind, has_prio, has_area, p0, has_proj,
+                               closed):
+    """needs_triage iff open and (missing kind/priority/area, or P0 without
+    a project event) — for every combination of label events."""
+    from code_intelligence_amd.triage.triage import TriageInfo
+    events = []
+    t = "2024-01-0{}T00:00:00Z"
+    if has_kind:
+        events.append({"__typename": "LabeledEvent", "createdAt": t.format(1),
+                       "label": {"name": "kind/bug"}})
+    prio_name = "priority/p0" if p0 else "priority/p2"
+    if has_prio:
+        events.append({"__typename": "LabeledEvent", "createdAt": t.format(2),
+                       "label": {"name": prio_name}})
+    if has_area:
+        events.append({"__typename": "LabeledEvent", "createdAt": t.format(3),
+                       "label": {"name": "area/ops"}})
+    if has_proj:
+        events.append({"__typename": "AddedToProjectEvent",
+                       "createdAt": t.format(4)})
+    issue = {"id": "x", "number": 1, "state": "CLOSED" if closed else "OPEN",
+             "closedAt": "2024-02-01T00:00:00Z" if closed else None,
+             "labels": {"edges": ([{"node": {"name": prio_name}}]
+                                  if has_prio else [])},
+             "projectCards": {"edges": []},
+             "timelineItems": {"edges": [{"node": e} for e in events]}}
+    info = TriageInfo.from_issue(issue)
+    expected = (not closed) and (
+        not (has_kind and has_prio and has_area)
+        or (has_prio and p0 and not has_proj))
+    assert info.needs_triage == expected
+    if not info.needs_triage:
+        assert info.triaged_at is not None
