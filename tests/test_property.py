@@ -125,3 +125,13 @@ def test_triage_rules_property(has_kind, has_prio, has_area, p0, has_proj,
     assert info.needs_triage == expected
     if not info.needs_triage:
         assert info.triaged_at is not None
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.lists(st.sampled_from(["alpha", "beta", "gamma", "delta"]),
+                min_size=1, max_size=30))
+def test_vocab_roundtrip_property(tokens):
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+    v = Vocab(defaults_specials + ["alpha", "beta", "gamma", "delta"])
+    ids = v.numericalize(tokens)
+    assert v.textify(ids).split() == tokens  # in-vocab roundtrip is exact
