@@ -22,3 +22,11 @@ worker:
 
 clean:
 	rm -rf code_intelligence_amd/ops/_build code_intelligence_amd/ops/*.so
+
+# repeated-suite stability soak (flake hunting)
+soak:
+	for i in 1 2 3; do $(PY) -m pytest tests -q -m "not gpu" -p no:cacheprovider | tail -1; done
+
+# per-kernel microbenchmarks (GPU box)
+kernel-bench:
+	$(PY) scripts/kernel_bench.py
