@@ -15,6 +15,7 @@ from __future__ import annotations
 import json
 import logging
 import os
+import threading
 from pathlib import Path
 from typing import List, Optional, Sequence
 
@@ -63,6 +64,11 @@ class InferenceWrapper:
         self.emb_sz = self.encoder.emb_sz
         self.pad_idx = self.encoder.pad_token
         self._graphs = {}
+        # the encoder carries hidden state across forward calls; serialize
+        # encodes so a threaded caller gets correct (serialized) results
+        # instead of silently corrupted ones (reference serving is
+        # single-threaded, app.py:128 — this guards misuse)
+        self._encode_lock = threading.Lock()
         self.use_graphs = use_graphs and self.device.type == "cuda"
         if self.use_graphs and getattr(self.encoder, "qrnn", False):
             # hipGraph replay of the QRNN encoder memory-faults on ROCm 7.2
@@ -97,6 +103,11 @@ class InferenceWrapper:
     @torch.no_grad()
     def _encode_batch(self, ids: torch.Tensor, lengths: torch.Tensor) -> torch.Tensor:
         """ids: (B, T) padded; returns (B, 3*emb_sz) pooled fp32."""
+        with self._encode_lock:
+            return self._encode_batch_locked(ids, lengths)
+
+    def _encode_batch_locked(self, ids: torch.Tensor, lengths: torch.Tensor
+                             ) -> torch.Tensor:
         B, T = ids.shape
         if self.use_graphs:
             # pad (B, T) up to fixed buckets so each shape is captured once
