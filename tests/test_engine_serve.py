@@ -149,3 +149,25 @@ def test_serve_max_tokens_cap(tmp_path, monkeypatch):
     assert len(w.numericalize(long_text)) == 64
     emb = w.get_pooled_features(long_text)
     assert emb.shape == (1, 48)
+
+
+def test_concurrent_requests_serialized_correctly(tmp_path):
+    """The wrapper's hidden-state carry must not corrupt results when
+    called from multiple threads (encode lock)."""
+    import threading
+    w = _tiny_wrapper(tmp_path)
+    ref = w.get_pooled_features("w1 w2 w3")
+    errs = []
+
+    def hammer():
+        for _ in range(40):
+            if not torch.allclose(w.get_pooled_features("w1 w2 w3"), ref,
+                                  atol=1e-5):
+                errs.append(1)
+
+    ts = [threading.Thread(target=hammer) for _ in range(3)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs
