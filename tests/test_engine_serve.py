@@ -171,3 +171,35 @@ def test_concurrent_requests_serialized_correctly(tmp_path):
     for t in ts:
         t.join()
     assert not errs
+
+
+def test_wire_level_http_contract(tmp_path):
+    """Real HTTP server + requests client: the reference client recipe
+    (np.frombuffer(r.content, '<f4')) works over the wire, with the md5
+    header, /metrics and the batched /texts route."""
+    import threading
+    import time
+    import requests
+    from werkzeug.serving import make_server
+    from code_intelligence_amd.serve.app import create_app
+
+    w = _tiny_wrapper(tmp_path)
+    app = create_app(wrapper=w)
+    srv = make_server("127.0.0.1", 0, app)   # ephemeral port
+    port = srv.server_port
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        base = f"http://127.0.0.1:{port}"
+        assert requests.get(f"{base}/healthz", timeout=5).json() == {"ok": True}
+        r = requests.post(f"{base}/text",
+                          json={"title": "crash", "body": "w1 w2"}, timeout=10)
+        emb = np.frombuffer(r.content, dtype="<f4")
+        assert emb.shape == (48,) and np.isfinite(emb).all()
+        assert len(r.headers["X-Embedding-MD5"]) == 32
+        r2 = requests.post(f"{base}/texts", json={"documents": [
+            {"title": "a", "body": "w1"}, {"title": "b", "body": "w2"}]},
+            timeout=10)
+        assert r2.json()["shape"] == [2, 48]
+    finally:
+        srv.shutdown()
