@@ -12,7 +12,7 @@ import logging
 import re
 import threading
 import time
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import yaml
 from flask import Flask, jsonify, request

@@ -14,7 +14,7 @@ import datetime
 import json
 import uuid
 from pathlib import Path
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 
 def _now() -> str:

@@ -13,7 +13,6 @@ it. Metrics use sklearn (CPU-side, tiny)."""
 from __future__ import annotations
 
 import pickle
-from pathlib import Path
 from typing import Dict, List, Optional, Sequence
 
 import numpy as np

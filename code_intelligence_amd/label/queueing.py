@@ -17,7 +17,6 @@ from __future__ import annotations
 import json
 import queue
 import threading
-import time
 import uuid
 from dataclasses import dataclass, field
 from pathlib import Path

@@ -19,12 +19,12 @@ import io
 import os
 import logging
 import tempfile
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import yaml
 
-from ..engine.embeddings import CLASSIFIER_DIMS, get_all_issue_text
+from ..engine.embeddings import get_all_issue_text
 from ..engine.inference import InferenceWrapper
 from ..gh.gcs_util import ObjectStore, default_store
 from .mlp import MLPWrapper

@@ -13,7 +13,7 @@ from __future__ import annotations
 import logging
 from typing import List, Optional
 
-from ..gh.graphql import GraphQLClient, ShardWriter, unpack_and_split_nodes
+from ..gh.graphql import GraphQLClient, ShardWriter
 
 log = logging.getLogger(__name__)
 

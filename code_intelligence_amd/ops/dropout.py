@@ -9,7 +9,6 @@ epilogue (tracked as the K4-fusion item in SURVEY.md §2.4).
 """
 from __future__ import annotations
 
-import torch
 from torch import Tensor
 
 __all__ = ["variational_dropout"]

@@ -20,7 +20,6 @@ import hashlib
 import logging
 import os
 
-import numpy as np
 from flask import Flask, jsonify, request
 
 from ..engine.inference import InferenceWrapper

@@ -17,14 +17,11 @@ from __future__ import annotations
 
 import dataclasses
 import datetime
-import json
 import logging
 import os
-from pathlib import Path
 from typing import List, Optional
 
 from ..gh.graphql import GraphQLClient, ShardWriter, unpack_and_split_nodes
-from ..gh.util import parse_issue_url
 
 log = logging.getLogger(__name__)
 
