@@ -1,6 +1,5 @@
 """Coverage for CLIs, callbacks, schedules (table-driven, offline)."""
 import json
-import math
 
 import pytest
 import torch

@@ -1,6 +1,5 @@
 """Control-plane tests: registry, ModelSync reconcile, needs-sync server,
 chatbot label matching (reference: Go test techniques — fakes, goldens)."""
-import json
 
 import pytest
 import yaml

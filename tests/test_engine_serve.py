@@ -6,7 +6,7 @@ import torch
 from code_intelligence_amd.engine.inference import InferenceWrapper, save_artifacts
 from code_intelligence_amd.models.awd_lstm import AWDLSTM
 from code_intelligence_amd.serve.app import create_app
-from code_intelligence_amd.text.tokenizer import Tokenizer, Vocab, defaults_specials
+from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
 
 
 def _tiny_wrapper(tmp_path):

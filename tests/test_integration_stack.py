@@ -3,7 +3,6 @@ server (in-process) -> repo-specific model over HTTP contract -> predictor
 routing -> worker applies labels from a queued event. The whole L1-L5
 production path (SURVEY.md §1 data-flow) with only the GitHub REST calls
 faked."""
-import numpy as np
 import pytest
 import torch
 

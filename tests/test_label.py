@@ -9,7 +9,7 @@ import yaml
 from code_intelligence_amd.label.automl_model import AutoMLModel
 from code_intelligence_amd.label.combined_model import CombinedLabelModels
 from code_intelligence_amd.label.issue_label_predictor import IssueLabelPredictor
-from code_intelligence_amd.label.mlp import MLPHead, MLPWrapper
+from code_intelligence_amd.label.mlp import MLPWrapper
 from code_intelligence_amd.label.models import IssueLabelModel
 from code_intelligence_amd.label.queueing import LocalQueue
 from code_intelligence_amd.label.repo_config import RepoConfig

@@ -4,7 +4,6 @@ import io
 
 import pytest
 import torch
-import torch.nn.functional as F
 
 from code_intelligence_amd.models.awd_lstm import AWDLSTM, WeightDroppedLSTM
 from code_intelligence_amd.ops.lstm import lstm_forward

@@ -1,5 +1,4 @@
 """CPU reference-path tests for the op layer (pool, CE, AdamW, dropout)."""
-import pytest
 import torch
 import torch.nn.functional as F
 

@@ -1,8 +1,6 @@
 """Sweep runner + universal-model training tests."""
-import json
 
 import numpy as np
-import pytest
 
 from code_intelligence_amd.gh import bigquery
 from code_intelligence_amd.label.trainers import kind_targets, train_universal_model

@@ -1,7 +1,6 @@
 """Transfer-learning pipeline tests: bulk embeddings, repo MLP training
 pipeline, TransferTrainer step (CPU)."""
 import numpy as np
-import pytest
 import torch
 
 from code_intelligence_amd.engine.embeddings import (CLASSIFIER_DIMS,

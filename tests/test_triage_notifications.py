@@ -1,10 +1,8 @@
 """Triage rule engine + notifications tests (reference techniques:
 golden issue fixtures + recording fakes — SURVEY.md §4)."""
-import json
 
 import pytest
 
-from code_intelligence_amd.gh.graphql import GraphQLClient
 from code_intelligence_amd.notifications.notifications import (
     NotificationManager, download_issues, parse_issue_shards)
 from code_intelligence_amd.triage.triage import (IssueTriage, TriageInfo,
