@@ -136,6 +136,62 @@ def test_gradient_accumulation_no_sync(tmp_path, qrnn):
         assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
 
 
+def _worker_bf16(rank, out_path):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(WORLD),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29531",
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        m = _build_model().to(torch.bfloat16)
+        tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
+        x, y = _make_batches(seed=400 + rank)
+        tr.dist.prepare()
+        tr.loss_on_batch(x, y).backward()
+        tr.dist.finalize()
+        if rank == 0:
+            grads = {n: p.grad.clone() for n, p in m.named_parameters()
+                     if p.grad is not None}
+            torch.save(grads, out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_ddp_bf16_params_reduce_in_fp32(tmp_path):
+    """bf16 model: the bucket buffer is fp32, so the 2-rank sum happens in
+    fp32 and the result equals averaging each rank's bf16 grad in fp32
+    then casting back — not a bf16 wire sum."""
+    ctx = mp.get_context("spawn")
+    out = str(tmp_path / "gb.pt")
+    procs = [ctx.Process(target=_worker_bf16, args=(r, out))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=100)
+        assert p.exitcode == 0
+    dist_grads = torch.load(out, weights_only=True)
+    m = _build_model().to(torch.bfloat16)
+    tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=False)
+    acc = {}
+    for rank in range(WORLD):
+        for p_ in m.parameters():
+            p_.grad = None
+        m.reset()
+        x, y = _make_batches(seed=400 + rank)
+        tr.loss_on_batch(x, y).backward()
+        for n, p_ in m.named_parameters():
+            if p_.grad is not None:
+                acc[n] = acc.get(n, 0) + p_.grad.float() / WORLD
+    checked = 0
+    for n in set(acc) & set(dist_grads):
+        assert dist_grads[n].dtype == torch.bfloat16
+        assert torch.allclose(dist_grads[n].float(), acc[n],
+                              atol=1e-2, rtol=1e-2), n
+        checked += 1
+    assert checked > 5
+
+
 def _worker_ws4(rank, out_path):
     os.environ.update(RANK=str(rank), WORLD_SIZE="4",
                       MASTER_ADDR="127.0.0.1", MASTER_PORT="29523",
@@ -183,3 +239,71 @@ def test_ddp_world4_matches_average(tmp_path):
                 acc[n] = acc.get(n, 0) + p_.grad / 4
     for n in set(acc) & set(dist_grads):
         assert torch.allclose(dist_grads[n], acc[n], atol=1e-5), n
+
+
+def _worker_train4(rank, out_path):
+    os.environ.update(RANK=str(rank), WORLD_SIZE="4",
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29541",
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=4)
+    try:
+        m = _build_model()
+        tr = LMTrainer(m, TrainConfig(alpha=0, beta=0, one_cycle=False),
+                       distributed=True)
+        losses = []
+        for step in range(3):
+            x, y = _make_batches(seed=500 + rank * 100 + step)
+            losses.append(tr.train_step(x, y, lr=1e-3))
+        if rank == 0:
+            torch.save({"params": {n: p.detach().clone()
+                                   for n, p in m.named_parameters()},
+                        "losses": losses}, out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_ddp_world4_training_matches_single_process(tmp_path):
+    """3 full optimizer steps at world 4 == single process stepping on the
+    manually-averaged grads of the same 4 shards (VERDICT r1 next-round #1:
+    single-vs-multi loss equivalence through Adam, not just raw grads)."""
+    ctx = mp.get_context("spawn")
+    out = str(tmp_path / "t4.pt")
+    procs = [ctx.Process(target=_worker_train4, args=(r, out))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=150)
+        assert p.exitcode == 0
+    got = torch.load(out, weights_only=True)
+
+    m = _build_model()
+    tr = LMTrainer(m, TrainConfig(alpha=0, beta=0, one_cycle=False),
+                   distributed=False)
+    hidden = {r: None for r in range(4)}
+    for step in range(3):
+        tr.opt.zero_grad(set_to_none=True)
+        acc = {}
+        for rank in range(4):
+            for p_ in m.parameters():
+                p_.grad = None
+            # each rank carries its own hidden state across steps
+            if hidden[rank] is None:
+                m.reset()
+            else:
+                m.encoder.hidden = hidden[rank]
+            x, y = _make_batches(seed=500 + rank * 100 + step)
+            tr.loss_on_batch(x, y).backward()
+            hidden[rank] = m.encoder.hidden
+            for n, p_ in m.named_parameters():
+                if p_.grad is not None:
+                    acc[n] = acc.get(n, 0) + p_.grad / 4
+        for n, p_ in m.named_parameters():
+            p_.grad = acc.get(n)
+        for g in tr.opt.param_groups:
+            g["lr"] = 1e-3
+        tr.opt.step()
+    for n, p_ in m.named_parameters():
+        assert torch.allclose(got["params"][n], p_.detach(),
+                              atol=1e-5, rtol=1e-4), n

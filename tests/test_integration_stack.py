@@ -128,3 +128,29 @@ def test_bench_contract_cpu_fallback():
     assert out["scaling"] == "weak" and out["data"] == "synthetic"
     assert out["value"] > 0 and out["n_gpus"] == 1
     assert {"model", "global_batch", "seq_len", "parallelism"} <= set(out["config"])
+
+
+@pytest.mark.timeout(420)
+def test_bench_world8_torchrun_cpu():
+    """Launch bench.py EXACTLY like the driver's 8-GPU scale run
+    (torch.distributed.run, --nproc-per-node 8) but on CPU/gloo: proves
+    the rendezvous, env parsing, DDP bucketer world-8 wiring and the
+    rank-0-only JSON contract without a GPU node (VERDICT r1 #1)."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+    root = Path(__file__).resolve().parents[1]
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29771", str(root / "bench.py"),
+         "--gpus", "8", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=390)
+    assert r.returncode == 0, (r.stdout[-400:], r.stderr[-1200:])
+    lines = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout[-1000:]  # rank 0 only
+    out = json.loads(lines[0])
+    assert out["config"]["parallelism"] == "dp8"
+    assert out["config"]["global_batch"] == 2 * 8  # CPU-fallback bs=2
+    assert out["value"] > 0
