@@ -84,11 +84,12 @@ class InferenceWrapper:
 
     def numericalize(self, text: str) -> List[int]:
         ids = self.vocab.numericalize(self.tokenizer.process_text(text))
-        # opt-in serve-side length cap (CI_SERVE_MAX_TOKENS=N): a pathological
-        # multi-MB issue body otherwise runs unbounded recurrent timesteps.
-        # Default 0 = unlimited (reference parity: inference.py processes
-        # full documents).
-        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "0"))
+        # serve-side length cap (CI_SERVE_MAX_TOKENS=N, 0 = unlimited): a
+        # pathological multi-MB issue body otherwise runs unbounded
+        # recurrent timesteps on a single-threaded server. Default 10000
+        # tokens (~40 KB of text) — far beyond any real issue; the
+        # reference survived the same exposure only by having 9 replicas.
+        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "10000"))
         return ids[:cap] if cap > 0 else ids
 
     @staticmethod
@@ -147,7 +148,7 @@ class InferenceWrapper:
 
     def texts_to_embedding(self, texts: Sequence[str], bs: int = 100) -> np.ndarray:
         # batched tokenization (native GIL-released core underneath)
-        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "0"))
+        cap = int(os.environ.get("CI_SERVE_MAX_TOKENS", "10000"))
         tok_lists = self.tokenizer.process_all(list(texts))
         docs = []
         for toks in tok_lists:

@@ -6,6 +6,11 @@ MI355X runtime image, and GitHub App auth is control-path, so this module
 implements the minimum: PEM (PKCS#1 'RSA PRIVATE KEY' or PKCS#8
 'PRIVATE KEY') DER parsing and PKCS1-v1.5 SHA-256 signatures with plain
 modular exponentiation. Verified against `openssl dgst -sha256 -sign`.
+
+Caveat (documented, accepted): Python ``pow`` is not constant-time, so
+signing leaks timing information about the private exponent. This path
+only ever runs against an operator-supplied App key on the control path;
+when the ``cryptography`` package is available, prefer it.
 """
 from __future__ import annotations
 
@@ -34,7 +39,8 @@ def _read_tlv(data: bytes, i: int) -> Tuple[int, bytes, int]:
 
 def _read_int(data: bytes, i: int) -> Tuple[int, int]:
     tag, content, nxt = _read_tlv(data, i)
-    assert tag == 0x02, f"expected INTEGER, got tag {tag:#x}"
+    if tag != 0x02:
+        raise ValueError(f"malformed key: expected INTEGER, got tag {tag:#x}")
     return int.from_bytes(content, "big"), nxt
 
 
@@ -47,7 +53,8 @@ def parse_rsa_private_key_pem(pem: str) -> Dict[str, int]:
         raise ValueError("no PEM private key found")
     der = base64.b64decode(re.sub(r"\s", "", m.group(2)))
     tag, seq, _ = _read_tlv(der, 0)
-    assert tag == 0x30
+    if tag != 0x30:
+        raise ValueError("malformed key: outer DER tag is not SEQUENCE")
     # PKCS#8 wraps a PKCS#1 blob inside an OCTET STRING after an
     # AlgorithmIdentifier; PKCS#1 starts with INTEGER version then n.
     i = 0
@@ -56,9 +63,11 @@ def parse_rsa_private_key_pem(pem: str) -> Dict[str, int]:
     if nxt_tag == 0x30:  # PKCS#8: AlgorithmIdentifier SEQUENCE
         _, _, i = _read_tlv(seq, i)          # skip algorithm id
         tag, octet, _ = _read_tlv(seq, i)    # OCTET STRING with PKCS#1
-        assert tag == 0x04
+        if tag != 0x04:
+            raise ValueError("malformed PKCS#8 key: expected OCTET STRING")
         tag, seq, _ = _read_tlv(octet, 0)
-        assert tag == 0x30
+        if tag != 0x30:
+            raise ValueError("malformed PKCS#8 key: inner tag is not SEQUENCE")
         i = 0
         version, i = _read_int(seq, i)
     n, i = _read_int(seq, i)

@@ -33,12 +33,11 @@ class FusedAdamW(torch.optim.Optimizer):
             lr = group["lr"]
             beta1, beta2 = group["betas"]
             eps, wd = group["eps"], group["weight_decay"]
-            params: List[Tensor] = []
-            grads: List[Tensor] = []
-            masters: List[Tensor] = []
-            exp_avgs: List[Tensor] = []
-            exp_avg_sqs: List[Tensor] = []
-            step_t = 0
+            # Bias correction depends on each param's own step count, which
+            # can diverge when grads are intermittently None (gradual
+            # unfreezing). Batch the fused launch per step-count — in
+            # steady state that is one launch, exactly as before.
+            by_step: dict = {}
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -50,28 +49,29 @@ class FusedAdamW(torch.optim.Optimizer):
                     state["exp_avg"] = torch.zeros_like(state["master"])
                     state["exp_avg_sq"] = torch.zeros_like(state["master"])
                 state["step"] += 1
-                step_t = state["step"]
-                params.append(p)
-                grads.append(p.grad)
-                masters.append(state["master"])
-                exp_avgs.append(state["exp_avg"])
-                exp_avg_sqs.append(state["exp_avg_sq"])
-            if not params:
-                continue
-            bc1 = 1 - beta1 ** step_t
-            bc2 = 1 - beta2 ** step_t
-            if params[0].is_cuda:
-                lib = ext.require()
-                lib.fused_adamw(params, grads, masters, exp_avgs, exp_avg_sqs,
-                                lr, beta1, beta2, eps, wd, bc1, bc2)
-            else:
-                for p, g, m, ea, eas in zip(params, grads, masters, exp_avgs, exp_avg_sqs):
-                    gf = g.to(torch.float32)
-                    m.mul_(1 - lr * wd)
-                    ea.mul_(beta1).add_(gf, alpha=1 - beta1)
-                    eas.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
-                    denom = (eas / bc2).sqrt_().add_(eps)
-                    m.addcdiv_(ea / bc1, denom, value=-lr)
-                    if m.data_ptr() != p.data_ptr():
-                        p.copy_(m.to(p.dtype))
+                by_step.setdefault(state["step"], []).append((p, state))
+            for step_t, entries in by_step.items():
+                params: List[Tensor] = [p for p, _ in entries]
+                grads: List[Tensor] = [p.grad for p, _ in entries]
+                masters = [s["master"] for _, s in entries]
+                exp_avgs = [s["exp_avg"] for _, s in entries]
+                exp_avg_sqs = [s["exp_avg_sq"] for _, s in entries]
+                bc1 = 1 - beta1 ** step_t
+                bc2 = 1 - beta2 ** step_t
+                if params[0].is_cuda:
+                    lib = ext.require()
+                    lib.fused_adamw(params, grads, masters, exp_avgs,
+                                    exp_avg_sqs, lr, beta1, beta2, eps, wd,
+                                    bc1, bc2)
+                else:
+                    for p, g, m, ea, eas in zip(params, grads, masters,
+                                                exp_avgs, exp_avg_sqs):
+                        gf = g.to(torch.float32)
+                        m.mul_(1 - lr * wd)
+                        ea.mul_(beta1).add_(gf, alpha=1 - beta1)
+                        eas.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+                        denom = (eas / bc2).sqrt_().add_(eps)
+                        m.addcdiv_(ea / bc1, denom, value=-lr)
+                        if m.data_ptr() != p.data_ptr():
+                            p.copy_(m.to(p.dtype))
         return loss

@@ -73,23 +73,21 @@ def _accum_grad(p: Tensor, g: Tensor) -> None:
         p.grad.add_(g.to(p.dtype))
 
 
-_fp8_cache: dict = {}
-
-
 def _fp8_weights(w_hh: Tensor):
-    """Per-row e4m3 quantization of the recurrent weight, cached by tensor
-    identity+version (serving weights are static). CI_SERVE_FP8W=1 halves
-    the weight stream that bounds small-batch serve latency."""
-    key = (id(w_hh), w_hh._version)
-    hit = _fp8_cache.get(key)
-    if hit is not None:
-        return hit
+    """Per-row e4m3 quantization of the recurrent weight (serving weights
+    are static). CI_SERVE_FP8W=1 halves the weight stream that bounds
+    small-batch serve latency. The cache lives ON the weight tensor
+    (instance attribute) so its lifetime is the tensor's own — an id()
+    keyed dict could hand back a stale quantization after the id is
+    recycled post-GC (round-1 advisor finding)."""
+    hit = getattr(w_hh, "_ci_fp8", None)
+    if hit is not None and hit[0] == w_hh._version:
+        return hit[1], hit[2]
     scale = w_hh.detach().abs().amax(dim=1, keepdim=True).float()         .clamp_min(1e-12) / 448.0
     q = (w_hh.detach().float() / scale).clamp(-448.0, 448.0)         .to(torch.float8_e4m3fn).view(torch.uint8).contiguous()
-    out = (q, scale.squeeze(1).contiguous())
-    _fp8_cache.clear()  # keep at most the live weight set small
-    _fp8_cache[key] = out
-    return out
+    out = (w_hh._version, q, scale.squeeze(1).contiguous())
+    w_hh._ci_fp8 = out
+    return out[1], out[2]
 
 
 def _cpu_lstm_loop(x: Tensor, h0: Tensor, c0: Tensor, w_ih: Tensor, w_hh: Tensor,

@@ -47,6 +47,10 @@ def _resolve_model_path(model_path: str | None) -> str:
 def create_app(wrapper: InferenceWrapper | None = None,
                model_path: str | None = None) -> Flask:
     app = Flask("issue_embedding_server")
+    # bound request bodies: one multi-MB body would otherwise occupy the
+    # single-threaded server for an unbounded stretch (advisor r1 finding)
+    app.config["MAX_CONTENT_LENGTH"] = int(
+        os.environ.get("CI_SERVE_MAX_BYTES", str(16 * 1024 * 1024)))
     if wrapper is None:
         wrapper = InferenceWrapper(
             model_path=_resolve_model_path(model_path),
