@@ -26,7 +26,8 @@ import torch
 from torch import Tensor, nn
 
 from ..ops.lstm import lstm_forward
-from ..ops.dropout import variational_dropout
+from ..ops.dropout import variational_dropout, dropconnect
+from ..ops.embedding import embedding_row_dropout
 
 __all__ = [
     "EmbeddingDropout",
@@ -91,6 +92,12 @@ class EmbeddingDropout(nn.Module):
         self.embed_p = embed_p
 
     def forward(self, words: Tensor, scale: Optional[float] = None) -> Tensor:
+        if words.is_cuda:
+            # K1 HIP kernel: row mask fused into the gather — the masked
+            # 60kx800 table is never materialized (embedding.hip)
+            return embedding_row_dropout(
+                self.emb.weight, words, self.embed_p, self.training,
+                self.emb.padding_idx, scale)
         if self.training and self.embed_p != 0:
             size = (self.emb.weight.size(0), 1)
             mask = self.emb.weight.new_empty(size).bernoulli_(1 - self.embed_p)
@@ -149,8 +156,10 @@ class WeightDroppedLSTM(nn.Module):
             self.module.weight_hh_l0.copy_(self.weight_hh_l0_raw)
 
     def _masked_weight(self) -> Tensor:
-        return nn.functional.dropout(
-            self.weight_hh_l0_raw, p=self.weight_p, training=self.training)
+        # K3: seeded DropConnect kernel on CUDA (no mask tensor, in-place
+        # grad masking); F.dropout reference on CPU (ops/dropout.py)
+        return dropconnect(self.weight_hh_l0_raw, self.weight_p,
+                           self.training)
 
     def forward(self, x: Tensor, state: Tuple[Tensor, Tensor]) -> Tuple[Tensor, Tuple[Tensor, Tensor]]:
         """x: (B, T, input_size); state: (h, c) each (B, hidden_size)."""

@@ -39,6 +39,8 @@ std::vector<std::vector<std::string>> tokenize_core_batch(
     const std::vector<std::string>& texts);
 at::Tensor emb_scatter(at::Tensor gout, at::Tensor ids, at::Tensor rowmask,
                        long V, long pad_idx);
+at::Tensor dropconnect_apply(at::Tensor w, long seed, double p);
+void dropconnect_grad_(at::Tensor g, long seed, double p);
 }  // namespace ci
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -65,4 +67,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tokenize_core_batch", &ci::tokenize_core_batch,
         "native ASCII tokenizer core (batch, GIL released)");
   m.def("emb_scatter", &ci::emb_scatter, "embedding grad scatter-add");
+  m.def("dropconnect_apply", &ci::dropconnect_apply,
+        "seeded DropConnect mask-scale (no mask tensor)");
+  m.def("dropconnect_grad_", &ci::dropconnect_grad_,
+        "in-place seeded DropConnect grad mask");
 }

@@ -36,6 +36,7 @@ SOURCES = [
     "adam.hip",
     "ce.hip",
     "embedding.hip",
+    "dropconnect.hip",
     "tokenizer.cpp",
 ]
 

@@ -259,3 +259,170 @@ def test_qrnn_layer_gpu_matches_cpu():
                       (c0.to(DEV, torch.bfloat16),) * 2)
     assert (out.float().cpu() - ref).abs().max() < 0.05
     assert (cT.float().cpu() - cT_ref).abs().max() < 0.05
+
+
+# ---- K1: embedding gather + row dropout (wired round 2) ------------------
+
+def test_emb_gather_scatter_matches_masked_embedding():
+    """emb_gather with an explicit row mask == F.embedding on the masked
+    table; emb_scatter == autograd's dW, both vs CPU fp32."""
+    from code_intelligence_amd.ops import extension as ext
+    lib = ext.require()
+    torch.manual_seed(3)
+    V, E, B, T, pad = 500, 64, 8, 12, 1
+    w = torch.randn(V, E)
+    ids = torch.randint(0, V, (B, T))
+    ids[0, :3] = pad
+    mask = (torch.rand(V) > 0.2).float() / 0.8
+
+    # CPU fp32 reference through autograd
+    w_ref = w.clone().requires_grad_(True)
+    out_ref = F.embedding(ids, w_ref * mask[:, None], padding_idx=pad)
+    g = torch.randn_like(out_ref)
+    out_ref.backward(g)
+
+    wg = w.to(DEV, torch.bfloat16)
+    out = lib.emb_gather(wg, ids.to(DEV).contiguous(), mask.to(DEV))
+    assert out.dtype == torch.bfloat16
+    assert torch.allclose(out.float().cpu(), out_ref.detach(), atol=0.03)
+    dw = lib.emb_scatter(g.to(DEV, torch.bfloat16), ids.to(DEV).contiguous(),
+                         mask.to(DEV), V, pad)
+    assert dw.dtype == torch.float32
+    assert torch.allclose(dw.cpu(), w_ref.grad, atol=0.05), \
+        (dw.cpu() - w_ref.grad).abs().max()
+
+
+def test_embedding_dropout_module_gpu_matches_cpu_eval():
+    """EmbeddingDropout eval path on GPU (kernel gather, no mask) must
+    equal the CPU F.embedding path bit-for-bit-ish in bf16 tolerance."""
+    from code_intelligence_amd.models.awd_lstm import EmbeddingDropout
+    torch.manual_seed(4)
+    emb = torch.nn.Embedding(300, 48, padding_idx=1)
+    mod = EmbeddingDropout(emb, 0.02).eval()
+    ids = torch.randint(0, 300, (4, 9))
+    ref = mod(ids)
+    out = EmbeddingDropout(emb.to(DEV).to(torch.bfloat16), 0.02).eval()(ids.to(DEV))
+    assert torch.allclose(out.float().cpu(), ref, atol=0.02)
+    # scale argument parity (serve path passes scale=None; engine uses it)
+    ref_s = mod(ids, scale=0.5)
+    out_s = EmbeddingDropout(emb.to(DEV).to(torch.bfloat16), 0.02).eval()(
+        ids.to(DEV), scale=0.5)
+    assert torch.allclose(out_s.float().cpu(), ref_s, atol=0.02)
+
+
+def test_embedding_dropout_module_gpu_train_grads():
+    """Training path: the fused-row-mask kernel's forward zeros whole word
+    rows consistently and backward routes masked grads to only the looked-
+    up rows (pad row excluded)."""
+    from code_intelligence_amd.models.awd_lstm import EmbeddingDropout
+    torch.manual_seed(5)
+    V, E, pad = 200, 32, 1
+    emb = torch.nn.Embedding(V, E, padding_idx=pad).to(DEV)
+    mod = EmbeddingDropout(emb, 0.3).train()
+    ids = torch.randint(2, V, (6, 7), device=DEV)
+    ids[0, 0] = pad
+    out = mod(ids)
+    # whole-row consistency: a dropped word is dropped at EVERY position
+    flat_ids = ids.reshape(-1)
+    flat_out = out.reshape(-1, E)
+    for tok in flat_ids.unique():
+        rows = flat_out[flat_ids == tok]
+        zeroed = (rows.abs().sum(dim=1) == 0)
+        assert bool(zeroed.all()) or bool((~zeroed).all())
+    loss = out.float().pow(2).sum()
+    loss.backward()
+    g = emb.weight.grad
+    assert g is not None
+    assert g[pad].abs().sum() == 0
+    looked = set(flat_ids.tolist()) - {pad}
+    untouched = [i for i in range(V) if i not in looked and i != pad]
+    assert g[untouched].abs().sum() == 0
+    assert g[sorted(looked)].abs().sum() > 0
+
+
+# ---- K3: seeded DropConnect kernel (wired round 2) -----------------------
+
+def test_dropconnect_kernel_stats_and_determinism():
+    from code_intelligence_amd.ops import extension as ext
+    lib = ext.require()
+    torch.manual_seed(6)
+    w = torch.randn(4 * 96, 96, device=DEV, dtype=torch.bfloat16)
+    p = 0.2
+    a = lib.dropconnect_apply(w, 12345, p)
+    b = lib.dropconnect_apply(w, 12345, p)
+    assert torch.equal(a, b)  # same seed -> same mask
+    c = lib.dropconnect_apply(w, 54321, p)
+    assert not torch.equal(a, c)
+    drop_frac = float((a == 0).float().mean())
+    assert abs(drop_frac - p) < 0.02
+    kept = a[a != 0]
+    ref = w[a != 0].float() / (1 - p)
+    assert torch.allclose(kept.float(), ref, atol=0.02)
+
+
+def test_dropconnect_autograd_grad_uses_same_mask():
+    from code_intelligence_amd.ops.dropout import _DropConnectFunction
+    torch.manual_seed(7)
+    w = torch.randn(64, 48, device=DEV, dtype=torch.float32,
+                    requires_grad=True)
+    out = _DropConnectFunction.apply(w, 0.25, 777)
+    g = torch.randn_like(out)
+    out.backward(g.clone())
+    dropped = out == 0
+    assert torch.all(w.grad[dropped] == 0)
+    assert torch.allclose(w.grad[~dropped], g[~dropped] / 0.75, atol=1e-6)
+
+
+def test_weightdrop_lstm_train_step_runs_with_kernels():
+    """A training forward+backward of WeightDroppedLSTM on GPU goes
+    through dropconnect_apply (train) and produces finite grads."""
+    from code_intelligence_amd.models.awd_lstm import WeightDroppedLSTM
+    torch.manual_seed(8)
+    m = WeightDroppedLSTM(32, 48, weight_p=0.2).to(DEV, torch.bfloat16).train()
+    x = torch.randn(4, 6, 32, device=DEV, dtype=torch.bfloat16)
+    h = torch.zeros(4, 48, device=DEV, dtype=torch.bfloat16)
+    out, _ = m(x, (h, h.clone()))
+    out.float().pow(2).mean().backward()
+    for n, p_ in m.named_parameters():
+        assert p_.grad is not None and torch.isfinite(p_.grad.float()).all(), n
+
+
+# ---- deployed-shape correctness (VERDICT r1 next-round #6) ---------------
+
+@pytest.mark.timeout(600)
+def test_lstm_deployed_shape_fused_vs_lib_vs_cpu_spot():
+    """H=2400, B=512, T=2 (the bench kernel grid incl. edge tiles): fused
+    and lib paths must agree with each other everywhere and with a CPU
+    fp32 reference on spot rows."""
+    from code_intelligence_amd.ops.lstm import lstm_forward, _cpu_lstm_loop
+    torch.manual_seed(9)
+    B, T, In, H = 512, 2, 800, 2400
+    x32 = torch.randn(B, T, In) * 0.5
+    w_ih = torch.randn(4 * H, In) * 0.02
+    w_hh = torch.randn(4 * H, H) * 0.02
+    b_ih = torch.randn(4 * H) * 0.02
+    b_hh = torch.randn(4 * H) * 0.02
+    spot = [0, 1, 255, 510, 511]
+    out_ref, h_ref, c_ref = _cpu_lstm_loop(
+        x32[spot], torch.zeros(len(spot), H), torch.zeros(len(spot), H),
+        w_ih, w_hh, b_ih, b_hh)
+
+    outs = {}
+    for mode in ("lib", "fused"):
+        os.environ["CI_LSTM_MODE"] = mode
+        args = [x32.to(DEV, torch.bfloat16),
+                torch.zeros(B, H, device=DEV, dtype=torch.bfloat16),
+                torch.zeros(B, H, device=DEV, dtype=torch.bfloat16),
+                w_ih.to(DEV, torch.bfloat16), w_hh.to(DEV, torch.bfloat16),
+                b_ih.to(DEV, torch.bfloat16), b_hh.to(DEV, torch.bfloat16)]
+        out, (hT, cT) = lstm_forward(*args)
+        outs[mode] = (out.float().cpu(), hT.float().cpu())
+    os.environ.pop("CI_LSTM_MODE", None)
+    # fused vs lib: same bf16 inputs, both fp32-accumulated -> tight
+    assert torch.allclose(outs["fused"][0], outs["lib"][0], atol=0.02), \
+        (outs["fused"][0] - outs["lib"][0]).abs().max()
+    # spot rows vs CPU fp32 (bf16 input rounding dominates)
+    for mode in ("lib", "fused"):
+        assert torch.allclose(outs[mode][0][spot], out_ref, atol=0.08), \
+            (mode, (outs[mode][0][spot] - out_ref).abs().max())
+        assert torch.allclose(outs[mode][1][spot], h_ref, atol=0.08), mode
