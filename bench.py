@@ -53,13 +53,88 @@ def parse_args():
                    help="bench the QRNN encoder variant (not the headline config)")
     p.add_argument("--lstm_mode", type=str, default=None,
                    help="override CI_LSTM_MODE (fused|lib)")
+    p.add_argument("--serve", action="store_true",
+                   help="serve-only mode: print the issue-embeddings/sec "
+                        "JSON line instead of the LM training line")
+    p.add_argument("--no-serve", action="store_true",
+                   help="skip the serve sub-benchmark in the LM line")
     return p.parse_args()
+
+
+def serve_bench(args, quick: bool = True) -> dict:
+    """Driver-timed serve metric (BASELINE.json 'issue-embeddings/sec
+    served', config 4): bulk batched path + single-request latency on the
+    deployed shape, synthetic issues, random-init weights."""
+    import numpy as np
+    from code_intelligence_amd.data.synthetic import synthetic_issue_texts
+    from code_intelligence_amd.engine.inference import InferenceWrapper
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+
+    torch.manual_seed(0)
+    on_gpu = torch.cuda.is_available()
+    emb, hid, layers, vocab = (args.emb, args.hid, args.layers, args.vocab) \
+        if on_gpu else (400, 400, 3, 30000)
+    words = [f"w{i}" for i in range(vocab - len(defaults_specials))]
+    v = Vocab(defaults_specials + words)
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    model = AWDLSTM(vocab_sz=len(v), emb_sz=emb, n_hid=hid, n_layers=layers,
+                    qrnn=args.qrnn)
+    w = InferenceWrapper(encoder=model.encoder, vocab=v)
+    n_bulk, n_single = (600, 30) if quick else (2000, 100)
+    if not on_gpu:
+        n_bulk, n_single = 40, 5
+    issues = synthetic_issue_texts(n_bulk, seed=3)
+    texts = [w.process_dict(d)["text"] for d in issues]
+    w.texts_to_embedding(texts[:min(128, n_bulk)], bs=100)  # warmup
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    out = w.texts_to_embedding(texts, bs=200)
+    if on_gpu:
+        torch.cuda.synchronize()
+    bulk_dt = time.perf_counter() - t0
+    lat = []
+    for d in issues[:n_single]:
+        t0 = time.perf_counter()
+        w.get_pooled_features(w.process_dict(d)["text"])
+        if on_gpu:
+            torch.cuda.synchronize()
+        lat.append(time.perf_counter() - t0)
+    lat.sort()
+    return {
+        "embeddings_per_sec_bulk": round(n_bulk / bulk_dt, 1),
+        "model": f"AWD-{'QRNN' if args.qrnn else 'LSTM'} {layers}x{hid} "
+                 f"emb{emb} vocab{vocab}",
+        "bulk_n": n_bulk, "bulk_bs": 200, "dim": int(out.shape[1]),
+        "single_p50_ms": round(lat[len(lat) // 2] * 1e3, 2),
+        "single_p95_ms": round(lat[min(len(lat) - 1,
+                                       int(len(lat) * 0.95))] * 1e3, 2),
+    }
 
 
 def main():
     args = parse_args()
     if args.lstm_mode:
         os.environ["CI_LSTM_MODE"] = args.lstm_mode
+    if args.serve:
+        s = serve_bench(args, quick=False)
+        print(json.dumps({
+            "metric": "issue-embeddings/sec served",
+            "value": s["embeddings_per_sec_bulk"],
+            "unit": "embeddings/s", "n_gpus": 1,
+            "steps": s["bulk_n"], "warmup": 128,
+            "ms_per_step": round(1000.0 / max(s["embeddings_per_sec_bulk"],
+                                              1e-9), 3),
+            "higher_is_better": True, "scaling": "weak",
+            "vs_baseline": None, "dtype": "bf16", "data": "synthetic",
+            "config": {
+                "model": s["model"],
+                "global_batch": s["bulk_bs"], "seq_len": "variable",
+                "parallelism": "serve-1gpu",
+                "single_p50_ms": s["single_p50_ms"],
+                "single_p95_ms": s["single_p95_ms"],
+            }}))
+        return
     rank, world = init_distributed()
     on_gpu = torch.cuda.is_available()
     if not on_gpu:
@@ -117,6 +192,18 @@ def main():
     n_gpus = world if on_gpu else args.gpus
     tokens = args.bs * args.seq * args.steps * world
     value = tokens / elapsed
+
+    serve = None
+    if rank == 0 and world == 1 and not args.no_serve:
+        # BASELINE.json's metric names BOTH halves ("LM tokens/sec ... and
+        # issue-embeddings/sec served"): attach a driver-timed serve
+        # measurement to the same JSON line. Free the
+        # training state first so the serve wrapper starts clean.
+        del trainer, model
+        if on_gpu:
+            torch.cuda.empty_cache()
+        serve = serve_bench(args, quick=True)
+
     if rank == 0:
         print(json.dumps({
             "metric": "LM tokens/sec (whole node)",
@@ -138,6 +225,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "device": "MI355X" if on_gpu else "cpu-fallback",
             },
+            **({"serve": serve} if serve is not None else {}),
         }))
 
 
