@@ -300,13 +300,14 @@ def test_embedding_dropout_module_gpu_matches_cpu_eval():
     emb = torch.nn.Embedding(300, 48, padding_idx=1)
     mod = EmbeddingDropout(emb, 0.02).eval()
     ids = torch.randint(0, 300, (4, 9))
+    # CPU references FIRST: .to(DEV) mutates the module in place
     ref = mod(ids)
-    out = EmbeddingDropout(emb.to(DEV).to(torch.bfloat16), 0.02).eval()(ids.to(DEV))
+    ref_s = mod(ids, scale=0.5)
+    gmod = EmbeddingDropout(emb.to(DEV).to(torch.bfloat16), 0.02).eval()
+    out = gmod(ids.to(DEV))
     assert torch.allclose(out.float().cpu(), ref, atol=0.02)
     # scale argument parity (serve path passes scale=None; engine uses it)
-    ref_s = mod(ids, scale=0.5)
-    out_s = EmbeddingDropout(emb.to(DEV).to(torch.bfloat16), 0.02).eval()(
-        ids.to(DEV), scale=0.5)
+    out_s = gmod(ids.to(DEV), scale=0.5)
     assert torch.allclose(out_s.float().cpu(), ref_s, atol=0.02)
 
 
