@@ -8,6 +8,8 @@
 // CDNA guide) so the (chunk, 60k) bias broadcast-add never materializes.
 #include "common.h"
 
+#include <hip/hip_fp8.h>
+
 namespace ci {
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -177,6 +179,210 @@ void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
           lse.data_ptr<float>(), scale.data_ptr<float>(), V);
     }
   });
+}
+
+// ---- fp8-resident logits variants (CI_CE_FP8R) ---------------------------
+// The (N, 60k) logits stay resident between forward and backward as OCP
+// e4m3 (half the HBM footprint and epilogue read traffic of bf16, and the
+// logits GEMM runs at the fp8 MFMA rate). Decode uses the gfx950
+// hardware cvt (packed fp8->f32 pairs at VALU rate, same idiom as
+// lstm_gemv.hip); encode uses __hip_fp8_e4m3 (hardware cvt on gfx950).
+// The logits scale arrives as a DEVICE pointer (updated from the running
+// lse max by the host between steps) so no host sync ever happens.
+
+template <int THREADS, bool HAS_BIAS>
+__global__ void ce_rowstats_fp8_kernel(const unsigned char* __restrict__ logits,
+                                       long row_stride,
+                                       const long* __restrict__ targets,
+                                       const float* __restrict__ bias,
+                                       const float* __restrict__ xscale_p,
+                                       float* __restrict__ lse,
+                                       float* __restrict__ tgt, int V) {
+  constexpr int NW = THREADS / kWave;
+  constexpr int VEC = 16;  // 16 fp8 per 16-B load
+  const int row = blockIdx.x;
+  const float xs = xscale_p[0];
+  const unsigned char* x = logits + (long)row * row_stride;
+  __shared__ float red_m[NW];
+  __shared__ float red_s[NW];
+  const int Vv = V / VEC * VEC;
+  float v16[VEC], b8[VEC];
+  float m = -3.4e38f, sum = 0.f;
+  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    const uint4 q = *reinterpret_cast<const uint4*>(x + v);
+    fp8x4_to_f32(q.x, v16);
+    fp8x4_to_f32(q.y, v16 + 4);
+    fp8x4_to_f32(q.z, v16 + 8);
+    fp8x4_to_f32(q.w, v16 + 12);
+    #pragma unroll
+    for (int e = 0; e < VEC; e += 8) {
+      if (HAS_BIAS) load_bias<8>(bias + v + e, b8);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k)
+        v16[e + k] = v16[e + k] * xs + (HAS_BIAS ? b8[k] : 0.f);
+    }
+    float bm = v16[0];
+    #pragma unroll
+    for (int e = 1; e < VEC; ++e) bm = fmaxf(bm, v16[e]);
+    if (bm > m) {
+      sum *= __expf(m - bm);
+      m = bm;
+    }
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) sum += __expf(v16[e] - m);
+  }
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
+    float lo[4];
+    fp8x4_to_f32((unsigned int)x[v], lo);  // low byte -> lo[0]
+    const float val = lo[0] * xs + (HAS_BIAS ? bias[v] : 0.f);
+    if (val > m) {
+      sum *= __expf(m - val);
+      m = val;
+    }
+    sum += __expf(val - m);
+  }
+  #pragma unroll
+  for (int off = kWave / 2; off > 0; off >>= 1) {
+    const float om = __shfl_down(m, off);
+    const float os = __shfl_down(sum, off);
+    const float nm = fmaxf(m, om);
+    sum = sum * __expf(m - nm) + os * __expf(om - nm);
+    m = nm;
+  }
+  if ((threadIdx.x & (kWave - 1)) == 0) {
+    red_m[threadIdx.x / kWave] = m;
+    red_s[threadIdx.x / kWave] = sum;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float gm = red_m[0];
+    for (int w = 1; w < NW; ++w) gm = fmaxf(gm, red_m[w]);
+    float gs = 0.f;
+    for (int w = 0; w < NW; ++w) gs += red_s[w] * __expf(red_m[w] - gm);
+    lse[row] = gm + __logf(gs);
+    float lo[4];
+    fp8x4_to_f32((unsigned int)x[targets[row]], lo);
+    tgt[row] = lo[0] * xs + (HAS_BIAS ? bias[targets[row]] : 0.f);
+  }
+}
+
+void ce_rowstats_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                     at::Tensor xscale, at::Tensor lse, at::Tensor tgt) {
+  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
+  TORCH_CHECK(logits.scalar_type() == at::ScalarType::Byte ||
+              logits.scalar_type() == at::ScalarType::Float8_e4m3fn,
+              "ce_rowstats_fp8: logits must be u8/e4m3");
+  const int N = logits.size(0), V = logits.size(1);
+  constexpr int THREADS = 256;
+  const bool hb = bias.numel() > 0;
+  if (hb) {
+    hipLaunchKernelGGL((ce_rowstats_fp8_kernel<THREADS, true>), dim3(N),
+        dim3(THREADS), 0, stream(),
+        reinterpret_cast<const unsigned char*>(logits.data_ptr()), (long)V,
+        targets.data_ptr<long>(), bias.data_ptr<float>(),
+        xscale.data_ptr<float>(), lse.data_ptr<float>(),
+        tgt.data_ptr<float>(), V);
+  } else {
+    hipLaunchKernelGGL((ce_rowstats_fp8_kernel<THREADS, false>), dim3(N),
+        dim3(THREADS), 0, stream(),
+        reinterpret_cast<const unsigned char*>(logits.data_ptr()), (long)V,
+        targets.data_ptr<long>(), nullptr,
+        xscale.data_ptr<float>(), lse.data_ptr<float>(),
+        tgt.data_ptr<float>(), V);
+  }
+}
+
+// in place: logits8 <- e4m3((softmax - onehot) * STORE) for the fp8 dh
+// GEMM, and scratch <- bf16((softmax - onehot) * dloss/N) for the bf16
+// dW GEMM + bias sum. One read of the fp8 row produces both.
+template <int THREADS, bool HAS_BIAS>
+__global__ void ce_dlogits_fp8_kernel(unsigned char* __restrict__ logits,
+                                      long row_stride,
+                                      const long* __restrict__ targets,
+                                      const float* __restrict__ bias,
+                                      const float* __restrict__ xscale_p,
+                                      const float* __restrict__ lse,
+                                      const float* __restrict__ scale,
+                                      __hip_bfloat16* __restrict__ scratch,
+                                      long srs, float store_scale, int V) {
+  constexpr int VEC = 16;
+  const int row = blockIdx.x;
+  const float xs = xscale_p[0];
+  const float l = lse[row];
+  const float sc = scale[0];
+  const int tgt = (int)targets[row];
+  unsigned char* x = logits + (long)row * row_stride;
+  __hip_bfloat16* s_out = scratch + (long)row * srs;
+  const int Vv = V / VEC * VEC;
+  float v16[VEC], b8[VEC];
+  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
+    const uint4 q = *reinterpret_cast<const uint4*>(x + v);
+    fp8x4_to_f32(q.x, v16);
+    fp8x4_to_f32(q.y, v16 + 4);
+    fp8x4_to_f32(q.z, v16 + 8);
+    fp8x4_to_f32(q.w, v16 + 12);
+    #pragma unroll
+    for (int e = 0; e < VEC; e += 8) {
+      if (HAS_BIAS) load_bias<8>(bias + v + e, b8);
+      #pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float p = __expf(v16[e + k] * xs + (HAS_BIAS ? b8[k] : 0.f) - l);
+        if (v + e + k == tgt) p -= 1.f;
+        v16[e + k] = p;
+      }
+    }
+    unsigned char q8[VEC];
+    __hip_bfloat16 s16[VEC];
+    #pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      q8[e] = __hip_fp8_e4m3(v16[e] * store_scale).__x;
+      s16[e] = __float2bfloat16(v16[e] * sc);
+    }
+    *reinterpret_cast<uint4*>(x + v) = *reinterpret_cast<const uint4*>(q8);
+    *reinterpret_cast<uint4*>(s_out + v) =
+        *reinterpret_cast<const uint4*>(s16);
+    *reinterpret_cast<uint4*>(s_out + v + 8) =
+        *reinterpret_cast<const uint4*>(s16 + 8);
+  }
+  for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
+    float lo[4];
+    fp8x4_to_f32((unsigned int)x[v], lo);
+    float p = __expf(lo[0] * xs + (HAS_BIAS ? bias[v] : 0.f) - l);
+    if (v == tgt) p -= 1.f;
+    x[v] = __hip_fp8_e4m3(p * store_scale).__x;
+    s_out[v] = __float2bfloat16(p * sc);
+  }
+}
+
+void ce_dlogits_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                    at::Tensor xscale, at::Tensor lse, at::Tensor scale,
+                    at::Tensor scratch, double store_scale) {
+  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits); CI_CHECK_CONTIG(scratch);
+  const int N = logits.size(0), V = logits.size(1);
+  TORCH_CHECK(scratch.size(0) >= N && scratch.size(1) == V,
+              "ce_dlogits_fp8: scratch too small");
+  TORCH_CHECK(scratch.scalar_type() == at::ScalarType::BFloat16);
+  constexpr int THREADS = 256;
+  const bool hb = bias.numel() > 0;
+  if (hb) {
+    hipLaunchKernelGGL((ce_dlogits_fp8_kernel<THREADS, true>), dim3(N),
+        dim3(THREADS), 0, stream(),
+        reinterpret_cast<unsigned char*>(logits.data_ptr()), (long)V,
+        targets.data_ptr<long>(), bias.data_ptr<float>(),
+        xscale.data_ptr<float>(), lse.data_ptr<float>(),
+        scale.data_ptr<float>(),
+        reinterpret_cast<__hip_bfloat16*>(scratch.data_ptr()),
+        (long)V, (float)store_scale, V);
+  } else {
+    hipLaunchKernelGGL((ce_dlogits_fp8_kernel<THREADS, false>), dim3(N),
+        dim3(THREADS), 0, stream(),
+        reinterpret_cast<unsigned char*>(logits.data_ptr()), (long)V,
+        targets.data_ptr<long>(), nullptr,
+        xscale.data_ptr<float>(), lse.data_ptr<float>(),
+        scale.data_ptr<float>(),
+        reinterpret_cast<__hip_bfloat16*>(scratch.data_ptr()),
+        (long)V, (float)store_scale, V);
+  }
 }
 
 }  // namespace ci

@@ -74,6 +74,16 @@ static __device__ __forceinline__ float sigmoidf_(float x) {
   return 1.0f / (1.0f + __expf(-x));
 }
 
+// OCP e4m3 unpack: gfx950 hardware cvt converts packed fp8 pairs to f32
+// at VALU rate (used by the fp8-weight GEMV and the fp8-resident CE).
+static __device__ __forceinline__ void fp8x4_to_f32(unsigned int u,
+                                                    float* out) {
+  typedef float f32x2_ __attribute__((ext_vector_type(2)));
+  f32x2_ lo = __builtin_amdgcn_cvt_pk_f32_fp8(u, false);
+  f32x2_ hi = __builtin_amdgcn_cvt_pk_f32_fp8(u, true);
+  out[0] = lo[0]; out[1] = lo[1]; out[2] = hi[0]; out[3] = hi[1];
+}
+
 // dispatch on an ATen dtype (float32 / bfloat16)
 #define CI_DISPATCH_FB(DTYPE, NAME, ...)                                   \
   [&] {                                                                    \

@@ -124,14 +124,7 @@ void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
 // shape); storing W as e4m3 with per-row scales halves that stream.
 // Hardware unpack: __builtin_amdgcn_cvt_pk_f32_fp8 converts packed fp8
 // pairs at VALU rate. Opt-in (CI_SERVE_FP8W=1), eval-path only.
-namespace {
-__device__ __forceinline__ void fp8x4_to_f32(unsigned int u, float* out) {
-  typedef float f32x2 __attribute__((ext_vector_type(2)));
-  f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(u, false);
-  f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(u, true);
-  out[0] = lo[0]; out[1] = lo[1]; out[2] = hi[0]; out[3] = hi[1];
-}
-}  // namespace
+// fp8x4_to_f32 unpack helper now lives in common.h (shared with ce.hip)
 
 __global__ __launch_bounds__(256) void lstm_cell_gemv_fp8(
     const __hip_bfloat16* __restrict__ h_prev, long h_rs,

@@ -427,3 +427,60 @@ def test_lstm_deployed_shape_fused_vs_lib_vs_cpu_spot():
         assert torch.allclose(outs[mode][0][spot], out_ref, atol=0.08), \
             (mode, (outs[mode][0][spot] - out_ref).abs().max())
         assert torch.allclose(outs[mode][1][spot], h_ref, atol=0.08), mode
+
+
+# ---- fp8-resident CE (CI_CE_FP8R, round 2) --------------------------------
+
+def _ce_both_paths(N, H, V, seed=11):
+    from code_intelligence_amd.ops.crossentropy import tied_decoder_ce
+    torch.manual_seed(seed)
+    h0 = torch.randn(N, H) * 0.5
+    w0 = torch.randn(V, H) * 0.05
+    b0 = torch.randn(V) * 0.01
+    t0 = torch.randint(0, V, (N,))
+    res = {}
+    for mode in ("bf16", "fp8r"):
+        os.environ["CI_CE_FP8R"] = "1" if mode == "fp8r" else "0"
+        h = h0.to(DEV, torch.bfloat16).requires_grad_(True)
+        w = w0.to(DEV, torch.bfloat16).requires_grad_(True)
+        b = b0.to(DEV, torch.bfloat16).requires_grad_(True)
+        loss = tied_decoder_ce(h, w, b, t0.to(DEV))
+        loss.backward()
+        res[mode] = (float(loss), h.grad.float().cpu(), w.grad.float().cpu(),
+                     b.grad.float().cpu())
+    os.environ.pop("CI_CE_FP8R", None)
+    return res
+
+
+@pytest.mark.timeout(300)
+def test_ce_fp8_resident_close_to_bf16():
+    """fp8-resident logits: loss within ~2% of the bf16-resident path and
+    gradients directionally identical (cosine > 0.98). Also implicitly
+    validates the _scaled_mm scale_result dequant convention — a wrong
+    convention would blow the loss up by orders of magnitude."""
+    res = _ce_both_paths(N=2048, H=160, V=4096)
+    lb, hb, wb, bb = res["bf16"]
+    lf, hf, wf, bf_ = res["fp8r"]
+    assert abs(lf - lb) / abs(lb) < 0.02, (lf, lb)
+
+    def cos(a, b):
+        return float((a.flatten() @ b.flatten()) /
+                     (a.norm() * b.norm() + 1e-30))
+    assert cos(hf, hb) > 0.98, cos(hf, hb)
+    assert cos(wf, wb) > 0.98, cos(wf, wb)
+    assert cos(bf_, bb) > 0.97, cos(bf_, bb)
+
+
+@pytest.mark.timeout(300)
+def test_ce_fp8_resident_tail_chunk():
+    """N not a multiple of 16 exercises the bf16-GEMM tail fallback for
+    both forward quantize-store and backward dh."""
+    os.environ["CI_CE_CHUNK"] = "64"
+    try:
+        res = _ce_both_paths(N=72 + 9, H=160, V=4096, seed=12)
+    finally:
+        os.environ.pop("CI_CE_CHUNK", None)
+    lb, hb, *_ = res["bf16"]
+    lf, hf, *_ = res["fp8r"]
+    assert abs(lf - lb) / abs(lb) < 0.03
+    assert torch.isfinite(hf).all()
