@@ -289,7 +289,11 @@ def tied_decoder_ce(h: Tensor, weight: Tensor, bias: Tensor | None,
     """h: (N, H) decoder input (already output-dropped); weight: (V, H) tied
     embedding; targets: (N,) int64. Returns scalar mean CE loss."""
     if h.is_cuda:
-        if _fp8r_enabled() and h.dtype == torch.bfloat16:
+        # fp8-resident only for TRAINING steps: evaluation (valid ppl) and
+        # any no-grad loss stays exact bf16, like fp8 training recipes
+        # that keep the eval loss in high precision.
+        if _fp8r_enabled() and h.dtype == torch.bfloat16 \
+                and torch.is_grad_enabled() and h.requires_grad:
             return _FusedCEFp8Function.apply(h, weight, bias, targets)
         return _FusedCEFunction.apply(h, weight, bias, targets)
     logits = torch.nn.functional.linear(h.float(), weight.float(),
