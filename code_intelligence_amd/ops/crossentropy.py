@@ -74,35 +74,28 @@ def _mm_maybe_fp8(a: Tensor, b_t: Tensor, out: Tensor,
 
 
 def _fp8r_enabled() -> bool:
-    """fp8-RESIDENT logits (CI_CE_FP8R): the (N, 60k) logits tile is
-    produced BY the fp8 MFMA GEMM directly in OCP e4m3 (scale_result /
-    out= into the resident buffer), the CE epilogues read/write e4m3, and
-    backward's dh GEMM runs fp8 too. Halves the CE GEMM FLOP-time and the
-    resident footprint (15.7 vs 31 GB at the bench shape). dW stays bf16
-    (dlog^T is column-major, which _scaled_mm rejects for mat1; the
-    epilogue kernel emits a bf16 scratch chunk for it in the same pass).
-    Target-logit values are computed exactly (bf16 gather-dot) so the
-    reported loss does not carry fp8 target error; lse is fp8-derived.
-    """
+    """fp8 CE GEMMs (CI_CE_FP8R): the logits GEMM runs with OCP e4m3
+    INPUTS (per-tensor scales) at the fp8 MFMA rate into the bf16-resident
+    buffer, and backward's dh GEMM runs fully fp8 from an e4m3 dlogits
+    scratch emitted by the dual epilogue kernel. dW stays bf16 (dlog^T is
+    column-major, which _scaled_mm rejects for mat1). Measured on MI355X
+    (scripts/fp8r_probe.py): fwd chunk GEMM 1.15 vs 1.59 ms, dh 0.78 vs
+    1.44 ms; a fully fp8-RESIDENT buffer was tried and LOSES because the
+    fp8-out GEMM epilogue is untuned (1.83 ms) and _scaled_mm ignores
+    scale_result on this stack. Logits carry fp8 input-quantization error
+    (~3.6% max rel, r1 probe); target logits are computed exactly via a
+    bf16 gather-dot and evaluation always runs the exact bf16 path."""
     return os.environ.get("CI_CE_FP8R", "0") == "1"
 
 
 _STORE = 448.0  # fixed dlogits store scale: |softmax - onehot| <= 1
-_AMAX: dict = {}  # device -> running logits-amax estimate (no host sync)
-
-
-def _amax_state(device) -> Tensor:
-    key = str(device)
-    if key not in _AMAX:
-        _AMAX[key] = torch.full((), 64.0, dtype=torch.float32, device=device)
-    return _AMAX[key]
 
 
 class _FusedCEFp8Function(torch.autograd.Function):
-    """fp8-resident variant of the fused tied-decoder CE (see
-    _fp8r_enabled). Numerics: logits carry e4m3 storage error (~6% rel);
-    measured against the bf16 path in tests/test_gpu_kernels.py and the
-    convergence check in profiles/BENCH_HISTORY.md."""
+    """fp8-GEMM variant of the fused tied-decoder CE (see _fp8r_enabled).
+    Loss/grad parity vs the bf16 path is locked by
+    tests/test_gpu_kernels.py (loss <2%, grad cosine >0.98) and the
+    markov convergence check in profiles/BENCH_HISTORY.md."""
 
     @staticmethod
     def forward(ctx, h: Tensor, weight: Tensor, bias: Tensor, targets: Tensor):
@@ -116,42 +109,39 @@ class _FusedCEFp8Function(torch.autograd.Function):
         tgt_logit = torch.empty(N, dtype=torch.float32, device=dev)
         b32 = bias.to(torch.float32) if bias is not None else _empty_f32(dev)
         tgt64 = targets.to(torch.int64)
-        # logits scale for e4m3 storage: running estimate of the positive
-        # logit max (lse bounds it within +log V), 2x margin. Stays on
-        # device — no sync. Saturation on the NEGATIVE side is harmless
-        # (exp -> 0); positive saturation is covered by the margin.
-        amax = _amax_state(dev)
-        s_out = (amax * (2.0 / 448.0)).clamp_min(16.0 / 448.0)
         sa = (h.detach().abs().amax().float() / 448.0).clamp_min(1e-12)
         sw = (weight.detach().abs().amax().float() / 448.0).clamp_min(1e-12)
         h8 = (h.detach() * (1.0 / sa)).clamp(-448., 448.).to(f8)
         w8 = (weight.detach() * (1.0 / sw)).clamp(-448., 448.).to(f8)
-        logits8 = torch.empty(N, V, dtype=f8, device=dev)
         w8_t = w8.t()  # (H, V) column-major view for mat2
         w_t = weight.t()
+        logits_full = None
+        try:
+            logits_full = torch.empty(N, V, dtype=h.dtype, device=dev)
+        except torch.cuda.OutOfMemoryError:
+            pass
+        scratch = None if logits_full is not None else \
+            torch.empty(min(C, N), V, dtype=h.dtype, device=dev)
         for s in range(0, N, C):
             e = min(N, s + C)
+            logits = logits_full[s:e] if logits_full is not None \
+                else scratch[: e - s]
             if (e - s) % 16 == 0 and H % 16 == 0 and V % 16 == 0:
                 torch._scaled_mm(h8[s:e], w8_t, scale_a=sa, scale_b=sw,
-                                 scale_result=s_out, out_dtype=f8,
-                                 out=logits8[s:e])
-            else:  # tail/odd shapes: bf16 GEMM then quantize
-                tmp = torch.mm(h[s:e], w_t)
-                logits8[s:e] = (tmp.float() / s_out).clamp(-448., 448.).to(f8)
-            lib.ce_rowstats_fp8(logits8[s:e], tgt64[s:e], b32, s_out,
-                                lse[s:e], tgt_logit[s:e])
+                                 out_dtype=h.dtype, out=logits)
+            else:  # tail/odd shapes: bf16 GEMM
+                torch.mm(h[s:e], w_t, out=logits)
+            lib.ce_rowstats(logits, tgt64[s:e], b32, lse[s:e],
+                            tgt_logit[s:e])
         # exact target logits (bf16 gather-dot, fp32 accumulation): the
-        # loss value must not carry fp8 target-logit error
+        # loss value must not carry fp8 input-quantization target error
         wrows = weight.detach()[tgt64]
         tgt_logit = (h.detach() * wrows).sum(dim=1, dtype=torch.float32)
         if bias is not None:
             tgt_logit = tgt_logit + b32[tgt64]
         loss = (lse - tgt_logit).mean()
-        # update the running amax estimate for the NEXT step
-        amax.copy_(torch.maximum(lse.detach().max(),
-                                 torch.full_like(amax, 16.0)))
-        ctx.save_for_backward(h, weight, b32, tgt64, lse, w8, sw, s_out)
-        ctx.logits8 = logits8
+        ctx.save_for_backward(h, weight, b32, tgt64, lse, w8, sw)
+        ctx.logits_full = logits_full
         ctx.has_bias = bias is not None
         ctx.bias_dtype = bias.dtype if bias is not None else None
         return loss
@@ -159,9 +149,10 @@ class _FusedCEFp8Function(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss: Tensor):
         lib = ext.require()
-        h, weight, b32, targets, lse, w8, sw, s_out = ctx.saved_tensors
-        logits8 = ctx.logits8
-        ctx.logits8 = None
+        f8 = torch.float8_e4m3fn
+        h, weight, b32, targets, lse, w8, sw = ctx.saved_tensors
+        logits_full = ctx.logits_full
+        ctx.logits_full = None
         has_bias = ctx.has_bias
         N, H = h.shape
         V = weight.shape[0]
@@ -172,24 +163,33 @@ class _FusedCEFp8Function(torch.autograd.Function):
         db = torch.zeros(V, dtype=torch.float32, device=dev) if has_bias else None
         scale = (dloss / N).to(torch.float32).reshape(1)
         sc_over_store = (scale.reshape(()) / _STORE)
-        scratch = torch.empty(min(C, N), V, dtype=h.dtype, device=dev)
+        scratch8 = torch.empty(min(C, N), V, dtype=f8, device=dev)
+        recompute = None if logits_full is not None else \
+            torch.empty(min(C, N), V, dtype=h.dtype, device=dev)
         # (V, H) column-major e4m3 weight for the fp8 dh GEMM
         w8_cm = w8.t().contiguous().t()
+        w_t = weight.t()
         bias_arg = b32 if has_bias else _empty_f32(dev)
         for s in range(0, N, C):
             e = min(N, s + C)
             c = e - s
-            # logits8[s:e] <- e4m3(q*448); scratch[:c] <- bf16(q*dloss/N)
-            lib.ce_dlogits_fp8(logits8[s:e], targets[s:e], bias_arg, s_out,
-                               lse[s:e], scale, scratch, _STORE)
+            if logits_full is not None:
+                dlog = logits_full[s:e]
+            else:
+                dlog = recompute[:c]
+                torch.mm(h[s:e], w_t, out=dlog)
+            # dlog <- bf16((softmax-onehot)*dloss/N) in place;
+            # scratch8 <- e4m3((softmax-onehot)*448) in the same read
+            lib.ce_dlogits_dual(dlog, targets[s:e], bias_arg, lse[s:e],
+                                scale, scratch8, _STORE)
             if c % 16 == 0 and H % 16 == 0 and V % 16 == 0:
-                torch._scaled_mm(logits8[s:e], w8_cm, scale_a=sc_over_store,
+                torch._scaled_mm(scratch8[:c], w8_cm, scale_a=sc_over_store,
                                  scale_b=sw, out_dtype=h.dtype, out=dh[s:e])
             else:
-                torch.mm(scratch[:c], weight, out=dh[s:e])
-            dw += torch.mm(scratch[:c].t(), h[s:e])
+                torch.mm(dlog, weight, out=dh[s:e])
+            dw += torch.mm(dlog.t(), h[s:e])
             if has_bias:
-                db += scratch[:c].sum(dim=0).to(torch.float32)
+                db += dlog.sum(dim=0).to(torch.float32)
         return (dh, dw.to(weight.dtype),
                 db.to(ctx.bias_dtype) if has_bias else None, None)
 

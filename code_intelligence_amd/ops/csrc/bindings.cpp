@@ -29,11 +29,9 @@ void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor bias,
                  at::Tensor lse, at::Tensor tgt);
 void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
                 at::Tensor lse, at::Tensor scale);
-void ce_rowstats_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
-                     at::Tensor xscale, at::Tensor lse, at::Tensor tgt);
-void ce_dlogits_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
-                    at::Tensor xscale, at::Tensor lse, at::Tensor scale,
-                    at::Tensor scratch, double store_scale);
+void ce_dlogits_dual(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                     at::Tensor lse, at::Tensor scale, at::Tensor scratch8,
+                     double store_scale);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                  std::vector<at::Tensor> masters, std::vector<at::Tensor> eas_,
                  std::vector<at::Tensor> eass, double lr, double b1, double b2,
@@ -66,10 +64,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "QRNN fo-pool backward scan (pre-activation gate grads)");
   m.def("ce_rowstats", &ci::ce_rowstats, "CE row logsumexp + target logit");
   m.def("ce_dlogits", &ci::ce_dlogits, "in-place (softmax-onehot)*scale");
-  m.def("ce_rowstats_fp8", &ci::ce_rowstats_fp8,
-        "CE row logsumexp + target logit over fp8-resident logits");
-  m.def("ce_dlogits_fp8", &ci::ce_dlogits_fp8,
-        "in-place fp8 dlogits + bf16 scratch from fp8-resident logits");
+  m.def("ce_dlogits_dual", &ci::ce_dlogits_dual,
+        "in-place bf16 dlogits + e4m3 scratch copy for the fp8 dh GEMM");
   m.def("fused_adamw", &ci::fused_adamw, "fused AdamW step");
   m.def("emb_gather", &ci::emb_gather, "embedding gather with row dropout");
   m.def("tokenize_core", &ci::tokenize_core, "native ASCII tokenizer core");

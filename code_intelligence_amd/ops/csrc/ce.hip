@@ -181,208 +181,94 @@ void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
   });
 }
 
-// ---- fp8-resident logits variants (CI_CE_FP8R) ---------------------------
-// The (N, 60k) logits stay resident between forward and backward as OCP
-// e4m3 (half the HBM footprint and epilogue read traffic of bf16, and the
-// logits GEMM runs at the fp8 MFMA rate). Decode uses the gfx950
-// hardware cvt (packed fp8->f32 pairs at VALU rate, same idiom as
-// lstm_gemv.hip); encode uses __hip_fp8_e4m3 (hardware cvt on gfx950).
-// The logits scale arrives as a DEVICE pointer (updated from the running
-// lse max by the host between steps) so no host sync ever happens.
-
-template <int THREADS, bool HAS_BIAS>
-__global__ void ce_rowstats_fp8_kernel(const unsigned char* __restrict__ logits,
-                                       long row_stride,
+// ---- fp8 CE GEMM support (CI_CE_FP8R) ------------------------------------
+// Measured on MI355X (scripts/fp8r_probe.py, profiles/fp8r_probe.log):
+// the fp8->fp8 GEMM epilogue is UNTUNED on this stack (1.83 ms vs 1.15 ms
+// for fp8->bf16 at the chunk shape) and _scaled_mm ignores scale_result,
+// so a fully fp8-RESIDENT logits buffer loses. The winning composition is
+// hybrid: bf16-resident logits produced by the fp8-input GEMM, and this
+// dual epilogue that (in one read of the row) rewrites the bf16 buffer to
+// (softmax-onehot)*dloss/N for the bf16 dW GEMM while emitting an e4m3
+// copy scaled by STORE=448 for the fp8 dh GEMM (0.78 vs 1.44 ms).
+template <typename T, int THREADS, bool HAS_BIAS>
+__global__ void ce_dlogits_dual_kernel(T* __restrict__ logits, long row_stride,
                                        const long* __restrict__ targets,
                                        const float* __restrict__ bias,
-                                       const float* __restrict__ xscale_p,
-                                       float* __restrict__ lse,
-                                       float* __restrict__ tgt, int V) {
-  constexpr int NW = THREADS / kWave;
-  constexpr int VEC = 16;  // 16 fp8 per 16-B load
+                                       const float* __restrict__ lse,
+                                       const float* __restrict__ scale,
+                                       unsigned char* __restrict__ scratch8,
+                                       long srs, float store_scale, int V) {
+  constexpr int VEC = 16 / sizeof(T);
   const int row = blockIdx.x;
-  const float xs = xscale_p[0];
-  const unsigned char* x = logits + (long)row * row_stride;
-  __shared__ float red_m[NW];
-  __shared__ float red_s[NW];
-  const int Vv = V / VEC * VEC;
-  float v16[VEC], b8[VEC];
-  float m = -3.4e38f, sum = 0.f;
-  for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
-    const uint4 q = *reinterpret_cast<const uint4*>(x + v);
-    fp8x4_to_f32(q.x, v16);
-    fp8x4_to_f32(q.y, v16 + 4);
-    fp8x4_to_f32(q.z, v16 + 8);
-    fp8x4_to_f32(q.w, v16 + 12);
-    #pragma unroll
-    for (int e = 0; e < VEC; e += 8) {
-      if (HAS_BIAS) load_bias<8>(bias + v + e, b8);
-      #pragma unroll
-      for (int k = 0; k < 8; ++k)
-        v16[e + k] = v16[e + k] * xs + (HAS_BIAS ? b8[k] : 0.f);
-    }
-    float bm = v16[0];
-    #pragma unroll
-    for (int e = 1; e < VEC; ++e) bm = fmaxf(bm, v16[e]);
-    if (bm > m) {
-      sum *= __expf(m - bm);
-      m = bm;
-    }
-    #pragma unroll
-    for (int e = 0; e < VEC; ++e) sum += __expf(v16[e] - m);
-  }
-  for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
-    float lo[4];
-    fp8x4_to_f32((unsigned int)x[v], lo);  // low byte -> lo[0]
-    const float val = lo[0] * xs + (HAS_BIAS ? bias[v] : 0.f);
-    if (val > m) {
-      sum *= __expf(m - val);
-      m = val;
-    }
-    sum += __expf(val - m);
-  }
-  #pragma unroll
-  for (int off = kWave / 2; off > 0; off >>= 1) {
-    const float om = __shfl_down(m, off);
-    const float os = __shfl_down(sum, off);
-    const float nm = fmaxf(m, om);
-    sum = sum * __expf(m - nm) + os * __expf(om - nm);
-    m = nm;
-  }
-  if ((threadIdx.x & (kWave - 1)) == 0) {
-    red_m[threadIdx.x / kWave] = m;
-    red_s[threadIdx.x / kWave] = sum;
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    float gm = red_m[0];
-    for (int w = 1; w < NW; ++w) gm = fmaxf(gm, red_m[w]);
-    float gs = 0.f;
-    for (int w = 0; w < NW; ++w) gs += red_s[w] * __expf(red_m[w] - gm);
-    lse[row] = gm + __logf(gs);
-    float lo[4];
-    fp8x4_to_f32((unsigned int)x[targets[row]], lo);
-    tgt[row] = lo[0] * xs + (HAS_BIAS ? bias[targets[row]] : 0.f);
-  }
-}
-
-void ce_rowstats_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
-                     at::Tensor xscale, at::Tensor lse, at::Tensor tgt) {
-  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits);
-  TORCH_CHECK(logits.scalar_type() == at::ScalarType::Byte ||
-              logits.scalar_type() == at::ScalarType::Float8_e4m3fn,
-              "ce_rowstats_fp8: logits must be u8/e4m3");
-  const int N = logits.size(0), V = logits.size(1);
-  constexpr int THREADS = 256;
-  const bool hb = bias.numel() > 0;
-  if (hb) {
-    hipLaunchKernelGGL((ce_rowstats_fp8_kernel<THREADS, true>), dim3(N),
-        dim3(THREADS), 0, stream(),
-        reinterpret_cast<const unsigned char*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), bias.data_ptr<float>(),
-        xscale.data_ptr<float>(), lse.data_ptr<float>(),
-        tgt.data_ptr<float>(), V);
-  } else {
-    hipLaunchKernelGGL((ce_rowstats_fp8_kernel<THREADS, false>), dim3(N),
-        dim3(THREADS), 0, stream(),
-        reinterpret_cast<const unsigned char*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), nullptr,
-        xscale.data_ptr<float>(), lse.data_ptr<float>(),
-        tgt.data_ptr<float>(), V);
-  }
-}
-
-// in place: logits8 <- e4m3((softmax - onehot) * STORE) for the fp8 dh
-// GEMM, and scratch <- bf16((softmax - onehot) * dloss/N) for the bf16
-// dW GEMM + bias sum. One read of the fp8 row produces both.
-template <int THREADS, bool HAS_BIAS>
-__global__ void ce_dlogits_fp8_kernel(unsigned char* __restrict__ logits,
-                                      long row_stride,
-                                      const long* __restrict__ targets,
-                                      const float* __restrict__ bias,
-                                      const float* __restrict__ xscale_p,
-                                      const float* __restrict__ lse,
-                                      const float* __restrict__ scale,
-                                      __hip_bfloat16* __restrict__ scratch,
-                                      long srs, float store_scale, int V) {
-  constexpr int VEC = 16;
-  const int row = blockIdx.x;
-  const float xs = xscale_p[0];
+  T* x = logits + (long)row * row_stride;
+  unsigned char* s8 = scratch8 + (long)row * srs;
   const float l = lse[row];
   const float sc = scale[0];
   const int tgt = (int)targets[row];
-  unsigned char* x = logits + (long)row * row_stride;
-  __hip_bfloat16* s_out = scratch + (long)row * srs;
   const int Vv = V / VEC * VEC;
-  float v16[VEC], b8[VEC];
+  float v8[VEC], b8[VEC];
   for (int v = threadIdx.x * VEC; v < Vv; v += THREADS * VEC) {
-    const uint4 q = *reinterpret_cast<const uint4*>(x + v);
-    fp8x4_to_f32(q.x, v16);
-    fp8x4_to_f32(q.y, v16 + 4);
-    fp8x4_to_f32(q.z, v16 + 8);
-    fp8x4_to_f32(q.w, v16 + 12);
-    #pragma unroll
-    for (int e = 0; e < VEC; e += 8) {
-      if (HAS_BIAS) load_bias<8>(bias + v + e, b8);
+    load_vec<T, VEC>(x + v, v8);
+    if (HAS_BIAS) {
+      load_bias<VEC>(bias + v, b8);
       #pragma unroll
-      for (int k = 0; k < 8; ++k) {
-        float p = __expf(v16[e + k] * xs + (HAS_BIAS ? b8[k] : 0.f) - l);
-        if (v + e + k == tgt) p -= 1.f;
-        v16[e + k] = p;
-      }
+      for (int e = 0; e < VEC; ++e) v8[e] += b8[e];
     }
+    T buf[VEC];
     unsigned char q8[VEC];
-    __hip_bfloat16 s16[VEC];
     #pragma unroll
     for (int e = 0; e < VEC; ++e) {
-      q8[e] = __hip_fp8_e4m3(v16[e] * store_scale).__x;
-      s16[e] = __float2bfloat16(v16[e] * sc);
+      float p = __expf(v8[e] - l);
+      if (v + e == tgt) p -= 1.f;
+      st(buf + e, p * sc);
+      q8[e] = __hip_fp8_e4m3(p * store_scale).__x;
     }
-    *reinterpret_cast<uint4*>(x + v) = *reinterpret_cast<const uint4*>(q8);
-    *reinterpret_cast<uint4*>(s_out + v) =
-        *reinterpret_cast<const uint4*>(s16);
-    *reinterpret_cast<uint4*>(s_out + v + 8) =
-        *reinterpret_cast<const uint4*>(s16 + 8);
+    *reinterpret_cast<int4*>(x + v) = *reinterpret_cast<const int4*>(buf);
+    if constexpr (VEC == 8) {  // bf16 path: 8 fp8 bytes
+      *reinterpret_cast<uint2*>(s8 + v) = *reinterpret_cast<const uint2*>(q8);
+    } else {                   // fp32 path: 4 fp8 bytes
+      *reinterpret_cast<unsigned int*>(s8 + v) =
+          *reinterpret_cast<const unsigned int*>(q8);
+    }
   }
   for (int v = Vv + threadIdx.x; v < V; v += THREADS) {
-    float lo[4];
-    fp8x4_to_f32((unsigned int)x[v], lo);
-    float p = __expf(lo[0] * xs + (HAS_BIAS ? bias[v] : 0.f) - l);
+    float p = __expf(ld(x + v) + (HAS_BIAS ? bias[v] : 0.f) - l);
     if (v == tgt) p -= 1.f;
-    x[v] = __hip_fp8_e4m3(p * store_scale).__x;
-    s_out[v] = __float2bfloat16(p * sc);
+    st(x + v, p * sc);
+    s8[v] = __hip_fp8_e4m3(p * store_scale).__x;
   }
 }
 
-void ce_dlogits_fp8(at::Tensor logits, at::Tensor targets, at::Tensor bias,
-                    at::Tensor xscale, at::Tensor lse, at::Tensor scale,
-                    at::Tensor scratch, double store_scale) {
-  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits); CI_CHECK_CONTIG(scratch);
+void ce_dlogits_dual(at::Tensor logits, at::Tensor targets, at::Tensor bias,
+                     at::Tensor lse, at::Tensor scale, at::Tensor scratch8,
+                     double store_scale) {
+  CI_CHECK_CUDA(logits); CI_CHECK_CONTIG(logits); CI_CHECK_CONTIG(scratch8);
   const int N = logits.size(0), V = logits.size(1);
-  TORCH_CHECK(scratch.size(0) >= N && scratch.size(1) == V,
-              "ce_dlogits_fp8: scratch too small");
-  TORCH_CHECK(scratch.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(scratch8.size(0) >= N && scratch8.size(1) == V,
+              "ce_dlogits_dual: scratch too small");
+  TORCH_CHECK(scratch8.scalar_type() == at::ScalarType::Byte ||
+              scratch8.scalar_type() == at::ScalarType::Float8_e4m3fn);
   constexpr int THREADS = 256;
   const bool hb = bias.numel() > 0;
-  if (hb) {
-    hipLaunchKernelGGL((ce_dlogits_fp8_kernel<THREADS, true>), dim3(N),
-        dim3(THREADS), 0, stream(),
-        reinterpret_cast<unsigned char*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), bias.data_ptr<float>(),
-        xscale.data_ptr<float>(), lse.data_ptr<float>(),
-        scale.data_ptr<float>(),
-        reinterpret_cast<__hip_bfloat16*>(scratch.data_ptr()),
-        (long)V, (float)store_scale, V);
-  } else {
-    hipLaunchKernelGGL((ce_dlogits_fp8_kernel<THREADS, false>), dim3(N),
-        dim3(THREADS), 0, stream(),
-        reinterpret_cast<unsigned char*>(logits.data_ptr()), (long)V,
-        targets.data_ptr<long>(), nullptr,
-        xscale.data_ptr<float>(), lse.data_ptr<float>(),
-        scale.data_ptr<float>(),
-        reinterpret_cast<__hip_bfloat16*>(scratch.data_ptr()),
-        (long)V, (float)store_scale, V);
-  }
+  CI_DISPATCH_FB(logits.scalar_type(), "ce_dlogits_dual", [&] {
+    if (hb) {
+      hipLaunchKernelGGL((ce_dlogits_dual_kernel<scalar_t, THREADS, true>),
+          dim3(N), dim3(THREADS), 0, stream(),
+          reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), bias.data_ptr<float>(),
+          lse.data_ptr<float>(), scale.data_ptr<float>(),
+          reinterpret_cast<unsigned char*>(scratch8.data_ptr()),
+          (long)V, (float)store_scale, V);
+    } else {
+      hipLaunchKernelGGL((ce_dlogits_dual_kernel<scalar_t, THREADS, false>),
+          dim3(N), dim3(THREADS), 0, stream(),
+          reinterpret_cast<scalar_t*>(logits.data_ptr()), (long)V,
+          targets.data_ptr<long>(), nullptr,
+          lse.data_ptr<float>(), scale.data_ptr<float>(),
+          reinterpret_cast<unsigned char*>(scratch8.data_ptr()),
+          (long)V, (float)store_scale, V);
+    }
+  });
 }
 
 }  // namespace ci

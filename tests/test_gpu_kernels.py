@@ -429,7 +429,7 @@ def test_lstm_deployed_shape_fused_vs_lib_vs_cpu_spot():
         assert torch.allclose(outs[mode][1][spot], h_ref, atol=0.08), mode
 
 
-# ---- fp8-resident CE (CI_CE_FP8R, round 2) --------------------------------
+# ---- fp8 CE GEMM path (CI_CE_FP8R, round 2) -------------------------------
 
 def _ce_both_paths(N, H, V, seed=11):
     from code_intelligence_amd.ops.crossentropy import tied_decoder_ce
@@ -454,10 +454,8 @@ def _ce_both_paths(N, H, V, seed=11):
 
 @pytest.mark.timeout(300)
 def test_ce_fp8_resident_close_to_bf16():
-    """fp8-resident logits: loss within ~2% of the bf16-resident path and
-    gradients directionally identical (cosine > 0.98). Also implicitly
-    validates the _scaled_mm scale_result dequant convention — a wrong
-    convention would blow the loss up by orders of magnitude."""
+    """fp8-input CE GEMMs + fp8 dh: loss within ~2% of the bf16 path and
+    gradients directionally identical (cosine > 0.98)."""
     res = _ce_both_paths(N=2048, H=160, V=4096)
     lb, hb, wb, bb = res["bf16"]
     lf, hf, wf, bf_ = res["fp8r"]
