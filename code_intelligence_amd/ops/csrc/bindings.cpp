@@ -32,6 +32,9 @@ void ce_dlogits(at::Tensor logits, at::Tensor targets, at::Tensor bias,
 void ce_dlogits_dual(at::Tensor logits, at::Tensor targets, at::Tensor bias,
                      at::Tensor lse, at::Tensor scale, at::Tensor scratch8,
                      double store_scale);
+at::Tensor artar_forward(at::Tensor out, at::Tensor r);
+std::vector<at::Tensor> artar_backward(at::Tensor out, at::Tensor r,
+                                       at::Tensor dloss, double ca, double cb);
 void fused_adamw(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
                  std::vector<at::Tensor> masters, std::vector<at::Tensor> eas_,
                  std::vector<at::Tensor> eass, double lr, double b1, double b2,
@@ -66,6 +69,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_dlogits", &ci::ce_dlogits, "in-place (softmax-onehot)*scale");
   m.def("ce_dlogits_dual", &ci::ce_dlogits_dual,
         "in-place bf16 dlogits + e4m3 scratch copy for the fp8 dh GEMM");
+  m.def("artar_forward", &ci::artar_forward,
+        "fused AR/TAR partial sums (one pass over out and r)");
+  m.def("artar_backward", &ci::artar_backward,
+        "fused AR/TAR gradient kernel (dout and dr in one pass)");
   m.def("fused_adamw", &ci::fused_adamw, "fused AdamW step");
   m.def("emb_gather", &ci::emb_gather, "embedding gather with row dropout");
   m.def("tokenize_core", &ci::tokenize_core, "native ASCII tokenizer core");

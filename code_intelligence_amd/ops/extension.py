@@ -37,6 +37,7 @@ SOURCES = [
     "ce.hip",
     "embedding.hip",
     "dropconnect.hip",
+    "artar.hip",
     "tokenizer.cpp",
 ]
 

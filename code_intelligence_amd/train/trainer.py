@@ -71,16 +71,12 @@ class LMTrainer:
         loss = tied_decoder_ce(out.reshape(-1, out.shape[-1]),
                                dec.decoder.weight, dec.decoder.bias,
                                y.reshape(-1))
-        if self.model.training:
-            if self.cfg.alpha:  # activation regularization on dropped output
-                # bf16 squares with fp32 accumulation: avoids materializing
-                # an fp32 copy of the (B,T,H) activations
-                loss = loss + self.cfg.alpha * out.pow(2).mean(dtype=torch.float32)
-            if self.cfg.beta:   # temporal AR on raw output diffs
-                r = raw_outputs[-1]
-                if r.shape[1] > 1:
-                    loss = loss + self.cfg.beta * (
-                        (r[:, 1:] - r[:, :-1]).pow(2).mean(dtype=torch.float32))
+        if self.model.training and (self.cfg.alpha or self.cfg.beta):
+            # fused AR/TAR kernel: one pass over the (B,T,H) activations
+            # instead of ~6 eager elementwise/reduce kernels (ops/artar.py)
+            from ..ops.artar import artar_loss
+            loss = loss + artar_loss(out, raw_outputs[-1],
+                                     self.cfg.alpha, self.cfg.beta)
         return loss
 
     def train_step(self, x, y, lr: float, mom: Optional[float] = None,
