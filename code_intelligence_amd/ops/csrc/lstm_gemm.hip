@@ -104,7 +104,15 @@ static __device__ __forceinline__ bf16x8 frag(const __hip_bfloat16* lds,
   return *reinterpret_cast<const bf16x8*>(lds + swz(row, col));
 }
 
-template <int BM, int BN>
+// counted wait: vmcnt <= n outstanding VMEM ops, all LDS ops drained
+// (lgkmcnt=0 keeps reg-staged edge tiles correct), expcnt unconstrained.
+// CDNA s_waitcnt imm: vmcnt[3:0]+[15:14], expcnt[6:4], lgkmcnt[13:8].
+template <int N>
+static __device__ __forceinline__ void waitcnt_vm() {
+  __builtin_amdgcn_s_waitcnt((0x7 << 4) | (N & 0xF) | (((N >> 4) & 0x3) << 14));
+}
+
+template <int BM, int BN, int STAGES = 2>
 __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
     const __hip_bfloat16* __restrict__ h_prev, long h_rs,
     const __hip_bfloat16* __restrict__ w_hh,   // (4H, H) checkpoint layout
@@ -157,27 +165,59 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
     }
   };
 
-  stage(0, 0);
-  __syncthreads();
-
-  for (int kt = 0; kt < nk; ++kt) {
-    const int cur = kt & 1;
-    if (kt + 1 < nk) stage(kt + 1, cur ^ 1);  // async: flight hides under MFMA
-    #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      bf16x8 af[4], bfr[2];
+  if constexpr (STAGES == 2) {
+    stage(0, 0);
+    __syncthreads();
+    for (int kt = 0; kt < nk; ++kt) {
+      const int cur = kt & 1;
+      if (kt + 1 < nk) stage(kt + 1, cur ^ 1);  // flight hides under MFMA
       #pragma unroll
-      for (int f = 0; f < 4; ++f) af[f] = frag(LA(cur), wm + f * 16, ks, lane);
-      #pragma unroll
-      for (int f = 0; f < 2; ++f) bfr[f] = frag(LB(cur), wn + f * 16, ks, lane);
-      #pragma unroll
-      for (int fm = 0; fm < 4; ++fm)
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 af[4], bfr[2];
         #pragma unroll
-        for (int fn = 0; fn < 2; ++fn)
-          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+        for (int f = 0; f < 4; ++f) af[f] = frag(LA(cur), wm + f * 16, ks, lane);
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) bfr[f] = frag(LB(cur), wn + f * 16, ks, lane);
+        #pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+          #pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+      }
+      __syncthreads();  // drains the glds queue (vmcnt0) + barrier
     }
-    __syncthreads();  // drains the glds queue (vmcnt0) + barrier
+  } else {
+    // 3-stage pipeline: prefetch depth 2 so a tile's glds flight hides
+    // under TWO MFMA phases (the 2-stage version stalls ~half the loop on
+    // load latency — PMC r1 'occupancy/latency-bound'). Counted
+    // s_waitcnt vmcnt(12) tolerates the 12 in-flight loads of the two
+    // younger tiles ((BM+BN)/32 = 6 glds per thread per stage).
+    stage(0, 0);
+    if (nk > 1) stage(1, 1);
+    if (nk > 2) stage(2, 2);
+    for (int kt = 0; kt < nk; ++kt) {
+      const int cur = kt % 3;
+      waitcnt_vm<12>();        // stage(kt) retired; kt+1/kt+2 may be in flight
+      __builtin_amdgcn_s_barrier();
+      #pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 af[4], bfr[2];
+        #pragma unroll
+        for (int f = 0; f < 4; ++f) af[f] = frag(LA(cur), wm + f * 16, ks, lane);
+        #pragma unroll
+        for (int f = 0; f < 2; ++f) bfr[f] = frag(LB(cur), wn + f * 16, ks, lane);
+        #pragma unroll
+        for (int fm = 0; fm < 4; ++fm)
+          #pragma unroll
+          for (int fn = 0; fn < 2; ++fn)
+            acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[fm], bfr[fn], acc[fm][fn], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_barrier();  // all waves done reading buf cur
+      if (kt + 3 < nk) stage(kt + 3, cur);  // reuse the freed slot
+    }
+    waitcnt_vm<0>();  // retire any tail-stage loads before the epilogue
   }
 
   // ---- epilogue: stash pre-activations in LDS, finish the cell ---------
@@ -226,7 +266,7 @@ __global__ __launch_bounds__(THREADS) void lstm_cell_fused(
 }
 
 // driver: whole-sequence forward, one fused launch per timestep.
-template <int BM, int BN>
+template <int BM, int BN, int STAGES = 2>
 static void run_fused(at::Tensor& xp, at::Tensor& bias, at::Tensor& h0,
                       at::Tensor& c0, at::Tensor& w_hh, at::Tensor& hs,
                       at::Tensor& cs, at::Tensor& gates) {
@@ -235,8 +275,17 @@ static void run_fused(at::Tensor& xp, at::Tensor& bias, at::Tensor& h0,
   const int MT = ceil_div(B, BM);
   const dim3 grid(MT * ceil_div(4 * H, BN));
   const size_t lds = std::max(
-      (size_t)2 * (BM + BN) * BK * sizeof(__hip_bfloat16),
+      (size_t)STAGES * (BM + BN) * BK * sizeof(__hip_bfloat16),
       (size_t)BM * BN * sizeof(float));
+  if (lds > 64 * 1024) {  // 3-stage config exceeds the default dyn-LDS cap
+    static bool attr_set = false;
+    if (!attr_set) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(&lstm_cell_fused<BM, BN, STAGES>),
+          hipFuncAttributeMaxDynamicSharedMemorySize, (int)lds);
+      attr_set = true;
+    }
+  }
   auto* hsp = reinterpret_cast<__hip_bfloat16*>(hs.data_ptr());
   auto* xpp = reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr());
   auto* gp = reinterpret_cast<__hip_bfloat16*>(gates.data_ptr());
@@ -247,8 +296,8 @@ static void run_fused(at::Tensor& xp, at::Tensor& bias, at::Tensor& h0,
     const __hip_bfloat16* hp = (t == 0) ? h0p : hsp + (long)(t - 1) * B * H;
     const float* cp = (t == 0) ? c0.data_ptr<float>()
                                : cs.data_ptr<float>() + (long)(t - 1) * B * H;
-    hipLaunchKernelGGL((lstm_cell_fused<BM, BN>), grid, dim3(THREADS), lds,
-        stream(),
+    hipLaunchKernelGGL((lstm_cell_fused<BM, BN, STAGES>), grid, dim3(THREADS),
+        lds, stream(),
         hp, (long)H, wp, xpp + (long)t * B * 4 * H, (long)4 * H,
         bias.data_ptr<float>(), cp, (long)H,
         hsp + (long)t * B * H, (long)H,
@@ -267,10 +316,14 @@ void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   TORCH_CHECK(w_hh.is_contiguous(), "w_hh must be contiguous");
   TORCH_CHECK(w_hh.size(1) % 8 == 0, "H must be a multiple of 8");
   const char* tile = getenv("CI_FUSED_TILE");
+  const char* pipe = getenv("CI_FUSED_PIPE");
+  const bool p3 = (pipe == nullptr) || std::string(pipe) == "3";  // default
   if (tile && std::string(tile) == "64x64") {
-    run_fused<64, 64>(xp, bias, h0, c0, w_hh, hs, cs, gates);
+    if (p3) run_fused<64, 64, 3>(xp, bias, h0, c0, w_hh, hs, cs, gates);
+    else    run_fused<64, 64, 2>(xp, bias, h0, c0, w_hh, hs, cs, gates);
   } else {
-    run_fused<128, 64>(xp, bias, h0, c0, w_hh, hs, cs, gates);
+    if (p3) run_fused<128, 64, 3>(xp, bias, h0, c0, w_hh, hs, cs, gates);
+    else    run_fused<128, 64, 2>(xp, bias, h0, c0, w_hh, hs, cs, gates);
   }
 }
 
