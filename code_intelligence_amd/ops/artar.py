@@ -19,12 +19,26 @@ from . import extension as ext
 __all__ = ["artar_loss"]
 
 
+def _flat_base(t: Tensor):
+    """The AR term is layout-agnostic: any contiguous covering of the
+    elements works. TensorIterator keeps the dropout output in the LSTM's
+    time-major layout, so accept either orientation's contiguous base.
+    Returns (base, swapped) or (None, False)."""
+    if t.is_contiguous():
+        return t, False
+    tt = t.transpose(0, 1)
+    if tt.is_contiguous():
+        return tt, True
+    return None, False
+
+
 class _ARTARFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, out: Tensor, r: Tensor, alpha: float, beta: float):
         # r: (B, T, H) transpose view of contiguous (T, B, H) storage
         lib = ext.require()
         r_tm = r.transpose(0, 1)
+        out, ctx.out_swapped = _flat_base(out)
         acc = lib.artar_forward(out, r_tm)  # [sum sq, sum diff sq] fp32
         n = max(out.numel(), 1)
         m = max(r.shape[0] * (r.shape[1] - 1) * r.shape[2], 1)
@@ -40,7 +54,10 @@ class _ARTARFunction(torch.autograd.Function):
         out, r = ctx.saved_tensors
         ca, cb = ctx.coeffs
         d32 = dloss.detach().to(torch.float32).reshape(1).contiguous()
-        dout, dr_tm = lib.artar_backward(out, r.transpose(0, 1), d32, ca, cb)
+        ob, _ = _flat_base(out)
+        dout, dr_tm = lib.artar_backward(ob, r.transpose(0, 1), d32, ca, cb)
+        if ctx.out_swapped:
+            dout = dout.transpose(0, 1)
         return dout, dr_tm.transpose(0, 1), None, None
 
 
@@ -48,7 +65,7 @@ def artar_loss(out: Tensor, r: Tensor, alpha: float, beta: float) -> Tensor:
     """out: output-dropped activations (B, T, H); r: raw last-layer
     output (B, T, H, typically a transpose view of time-major storage).
     Returns the scalar regularization term (0-dim fp32 tensor)."""
-    if (out.is_cuda and out.is_contiguous() and out.dim() == 3
+    if (out.is_cuda and out.dim() == 3 and _flat_base(out)[0] is not None
             and r.dim() == 3 and r.transpose(0, 1).is_contiguous()
             and r.shape[1] > 1
             and (r.shape[0] * r.shape[2]) % 8 == 0
