@@ -80,7 +80,7 @@ def serve_bench(args, quick: bool = True) -> dict:
     model = AWDLSTM(vocab_sz=len(v), emb_sz=emb, n_hid=hid, n_layers=layers,
                     qrnn=args.qrnn)
     w = InferenceWrapper(encoder=model.encoder, vocab=v)
-    n_bulk, n_single = (600, 30) if quick else (2000, 100)
+    n_bulk, n_single = (1500, 30) if quick else (2000, 100)
     if not on_gpu:
         n_bulk, n_single = 40, 5
     issues = synthetic_issue_texts(n_bulk, seed=3)
