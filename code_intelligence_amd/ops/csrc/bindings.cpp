@@ -10,6 +10,10 @@ void lstm_seq_forward_lib(at::Tensor xp, at::Tensor bias, at::Tensor h0,
 void lstm_seq_forward_fused(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                             at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
                             at::Tensor cs, at::Tensor gates);
+void lstm_seq_forward_lib_fp8(at::Tensor xp, at::Tensor bias, at::Tensor h0,
+                              at::Tensor c0, at::Tensor w8, at::Tensor wscale,
+                              at::Tensor hs, at::Tensor cs, at::Tensor gates);
+void quantize_e4m3(at::Tensor src, at::Tensor dst, at::Tensor scale);
 void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
                            at::Tensor cs, at::Tensor gates);
@@ -54,6 +58,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "LSTM sequence forward (hipBLASLt GEMM + pointwise cell kernel)");
   m.def("lstm_seq_forward_fused", &ci::lstm_seq_forward_fused,
         "LSTM sequence forward (fused MFMA cell kernel)");
+  m.def("lstm_seq_forward_lib_fp8", &ci::lstm_seq_forward_lib_fp8,
+        "LSTM sequence forward (fp8 recurrent scaled_mm + pointwise cell)");
+  m.def("quantize_e4m3", &ci::quantize_e4m3,
+        "one-pass e4m3 quantize: dst = e4m3(src/scale)");
   m.def("lstm_seq_forward_gemv", &ci::lstm_seq_forward_gemv,
         "LSTM sequence forward (fused GEMV+cell kernel, small batch)");
   m.def("lstm_seq_forward_gemv_fp8", &ci::lstm_seq_forward_gemv_fp8,

@@ -38,6 +38,7 @@ SOURCES = [
     "embedding.hip",
     "dropconnect.hip",
     "artar.hip",
+    "fp8util.hip",
     "tokenizer.cpp",
 ]
 

@@ -580,3 +580,59 @@ def test_artar_engages_on_trainer_layout():
     assert abs(float(reg) - float(ref)) / float(ref) < 0.02
     assert torch.allclose(out.grad.float(), out2.grad, atol=0.02)
     assert torch.allclose(r.grad.float(), r2.grad, atol=0.02)
+
+
+# ---- fp8 LSTM GEMM path (CI_LSTM_FP8, round 2) ----------------------------
+
+@pytest.mark.timeout(300)
+def test_lstm_fp8_close_to_bf16():
+    """fp8 input-projection + fp8 recurrent GEMM (fixed 1/448 h scale):
+    outputs within e4m3 tolerance of the bf16 lib path, grads
+    directionally identical."""
+    from code_intelligence_amd.ops.lstm import lstm_forward
+    torch.manual_seed(21)
+    B, T, In, H = 32, 12, 64, 96
+
+    def run(fp8):
+        os.environ["CI_LSTM_FP8"] = "1" if fp8 else "0"
+        torch.manual_seed(22)
+        x = (torch.randn(B, T, In) * 0.5).to(DEV, torch.bfloat16) \
+            .requires_grad_(True)
+        w_ih = (torch.randn(4 * H, In) * 0.1).to(DEV, torch.bfloat16) \
+            .requires_grad_(True)
+        w_hh = (torch.randn(4 * H, H) * 0.1).to(DEV, torch.bfloat16) \
+            .requires_grad_(True)
+        b = torch.zeros(4 * H, device=DEV, dtype=torch.bfloat16)
+        h0 = torch.zeros(B, H, device=DEV, dtype=torch.bfloat16)
+        out, (hT, cT) = lstm_forward(x, h0, h0.clone(), w_ih, w_hh, b,
+                                     b.clone())
+        out.float().pow(2).mean().backward()
+        return (out.detach().float().cpu(), x.grad.float().cpu(),
+                w_hh.grad.float().cpu())
+
+    try:
+        out_b, gx_b, gw_b = run(False)
+        out_f, gx_f, gw_f = run(True)
+    finally:
+        os.environ.pop("CI_LSTM_FP8", None)
+    # forward: fp8 quantization noise bounded
+    assert torch.allclose(out_f, out_b, atol=0.06), \
+        (out_f - out_b).abs().max()
+
+    def cos(a, b):
+        return float((a.flatten() @ b.flatten()) /
+                     (a.norm() * b.norm() + 1e-30))
+    assert cos(gx_f, gx_b) > 0.99, cos(gx_f, gx_b)
+    assert cos(gw_f, gw_b) > 0.99, cos(gw_f, gw_b)
+
+
+def test_quantize_e4m3_kernel():
+    from code_intelligence_amd.ops import extension as ext
+    lib = ext.require()
+    torch.manual_seed(23)
+    x = torch.randn(1000, device=DEV, dtype=torch.bfloat16) * 3
+    s = (x.abs().amax().float() / 448.0)
+    q = torch.empty(1000, dtype=torch.float8_e4m3fn, device=DEV)
+    lib.quantize_e4m3(x, q, s)
+    ref = ((x.float() / s).clamp(-448, 448)).to(torch.float8_e4m3fn)
+    assert torch.equal(q.view(torch.uint8), ref.view(torch.uint8))
