@@ -426,3 +426,24 @@ def test_spool_queue_durable_acks(tmp_path):
     del q2
     q3 = LocalQueue(spool_path=spool)  # everything acked now
     assert q3.pull(timeout=0.05) is None
+
+
+def test_shipped_universal_artifact_loads_and_separates():
+    """The committed model_files/universal artifact (trained by
+    scripts/make_universal_artifact.py, report in
+    docs/universal_model_report.md) loads and separates the three kinds
+    on obviously-typed issues."""
+    from pathlib import Path
+    from code_intelligence_amd.label.universal_kind_label_model import \
+        UniversalKindLabelModel
+    root = Path(__file__).resolve().parents[1] / "model_files" / "universal"
+    model = UniversalKindLabelModel.load(root)
+    bug = model.predict_issue_labels(
+        "o", "r", "crash with traceback", ["segfault error broken fails"])
+    feat = model.predict_issue_labels(
+        "o", "r", "add support for flag", ["implement option enhancement"])
+    q = model.predict_issue_labels(
+        "o", "r", "how to understand usage", ["question help why example"])
+    assert "kind/bug" in bug and "kind/feature" not in bug
+    assert "kind/feature" in feat
+    assert "kind/question" in q
