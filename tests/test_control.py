@@ -135,3 +135,30 @@ def test_chatbot_label_matching(tmp_path):
     r2 = c.post("/dialogflow/webhook", json={
         "queryResult": {"parameters": {"area": "nosuch"}}})
     assert "could not find" in r2.get_json()["fulfillmentText"]
+
+
+def test_k8s_manifests_parse_and_mirror_reference_topology():
+    """deploy/k8s base: valid YAML, /healthz readiness probe kept, worker
+    env contract (PROJECT/ISSUE_EVENT_*/MODEL_CONFIG) kept, 1-GPU
+    embedding server replaces the reference's 9 CPU replicas."""
+    import glob
+    from pathlib import Path
+    import yaml
+    root = Path(__file__).resolve().parents[1] / "deploy" / "k8s"
+    docs = []
+    for f in glob.glob(str(root / "**" / "*.yaml"), recursive=True):
+        docs += [d for d in yaml.safe_load_all(open(f)) if d]
+    kinds = {(d["kind"], d["metadata"]["name"]): d for d in docs
+             if "metadata" in d}
+    emb = kinds[("Deployment", "issue-embedding-server")]
+    c = emb["spec"]["template"]["spec"]["containers"][0]
+    assert c["resources"]["limits"]["amd.com/gpu"] == 1
+    assert c["readinessProbe"]["httpGet"]["path"] == "/healthz"
+    worker = kinds[("Deployment", "label-worker")]
+    assert worker["spec"]["replicas"] == 5
+    env = {e["name"] for e in
+           worker["spec"]["template"]["spec"]["containers"][0]["env"]}
+    assert {"PROJECT", "ISSUE_EVENT_TOPIC", "ISSUE_EVENT_SUBSCRIPTION",
+            "MODEL_CONFIG", "ISSUE_EMBEDDING_SERVICE"} <= env
+    assert ("Service", "issue-embedding-server") in kinds
+    assert ("Deployment", "modelsync") in kinds
