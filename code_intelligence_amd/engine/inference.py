@@ -70,13 +70,12 @@ class InferenceWrapper:
         # single-threaded, app.py:128 — this guards misuse)
         self._encode_lock = threading.Lock()
         self.use_graphs = use_graphs and self.device.type == "cuda"
-        if self.use_graphs and getattr(self.encoder, "qrnn", False):
-            # hipGraph replay of the QRNN encoder memory-faults on ROCm 7.2
-            # (capture of the chunked gate GEMM); eager QRNN serve is already
-            # launch-light, so the opt-in is ignored rather than risked.
-            log.warning("CI_SERVE_GRAPHS ignored for QRNN encoders "
-                        "(graph capture unsupported; serving eagerly)")
-            self.use_graphs = False
+        # (r1 gated QRNN graphs off after replay memory-faults; diagnosed
+        # in r2 as WeightDroppedQRNN reassigning prev_x to a capture-pool
+        # allocation — fixed by copying into a stable buffer, validated by
+        # scripts/qrnn_graph_repro.py rung 7. Graphs remain opt-in for
+        # both encoder families: bucket padding outweighs launch savings
+        # at measured shapes, profiles/BENCH_HISTORY.md.)
 
     # --- reference-parity helpers -----------------------------------------
     def process_dict(self, data: dict) -> dict:

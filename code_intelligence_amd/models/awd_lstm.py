@@ -216,7 +216,18 @@ class WeightDroppedQRNN(nn.Module):
         h, cT = self._qrnn_forward(x, state[1], w, self.bias,
                                    self.window, prev)
         if self.window == 2:
-            self.prev_x = x[:, -1:].detach()
+            nx = x[:, -1:].detach()
+            if (self.prev_x is not None and self.prev_x.shape == nx.shape
+                    and self.prev_x.device == nx.device
+                    and self.prev_x.dtype == nx.dtype):
+                # copy into the STABLE buffer instead of reassigning: a
+                # fresh allocation here inside a hipGraph capture lives in
+                # the capture-private pool and the python reassignment
+                # frees the captured pointer -> memory fault on replay
+                # (the r1 QRNN graph-capture fault, scripts/qrnn_graph_repro)
+                self.prev_x.copy_(nx)
+            else:
+                self.prev_x = nx.clone()
         return h, (h[:, -1], cT)
 
 
