@@ -157,17 +157,110 @@ class LocalBroker:
             LocalQueue(spool_path=str(spool)).publish(data, **attributes)
 
 
-def queue_from_env() -> BaseQueue:
+class PubSubRestQueue(BaseQueue):
+    """Cloud Pub/Sub adapter over the REST v1 surface with an INJECTABLE
+    transport — the python equivalent of the reference's RoundTripper
+    test seam (go/cmd/automl/pkg/client/client_test.go:18-31). The
+    google-cloud-pubsub SDK is not in this image; the wire protocol is
+    small enough to speak directly:
+
+      pull        POST .../subscriptions/{sub}:pull {"maxMessages": N}
+      acknowledge POST .../subscriptions/{sub}:acknowledge {"ackIds":[..]}
+      publish     POST .../topics/{topic}:publish {"messages":[{data:b64}]}
+      create      PUT  .../subscriptions/{sub} | .../topics/{topic}
+
+    ``transport(method, url, json_body, headers) -> (status, json_dict)``;
+    the default uses requests + a bearer-token provider. Contract tests
+    run against recorded wire fixtures (tests/test_label.py)."""
+
+    API = "https://pubsub.googleapis.com/v1"
+
+    def __init__(self, project: str, topic: str, subscription: str,
+                 transport: Optional[Callable] = None,
+                 token_provider: Optional[Callable[[], str]] = None):
+        self.project, self.topic, self.subscription = project, topic, subscription
+        self._transport = transport or self._requests_transport
+        self._token_provider = token_provider
+
+    # --- wire plumbing ----------------------------------------------------
+    def _requests_transport(self, method, url, body, headers):
+        import requests
+        r = requests.request(method, url, json=body, headers=headers,
+                             timeout=30)
+        return r.status_code, (r.json() if r.content else {})
+
+    def _call(self, method: str, path: str, body: dict) -> dict:
+        headers = {"Content-Type": "application/json"}
+        if self._token_provider is not None:
+            headers["Authorization"] = f"Bearer {self._token_provider()}"
+        status, payload = self._transport(
+            method, f"{self.API}/{path}", body, headers)
+        if status == 409:      # already-exists from idempotent creates
+            return payload
+        if status >= 400:
+            raise RuntimeError(f"pubsub {method} {path} -> {status}: {payload}")
+        return payload
+
+    # --- BaseQueue contract -----------------------------------------------
+    def publish(self, data: bytes = b"", **attributes) -> str:
+        import base64
+        payload = self._call(
+            "POST", f"projects/{self.project}/topics/{self.topic}:publish",
+            {"messages": [{
+                "data": base64.b64encode(data).decode(),
+                "attributes": {k: str(v) for k, v in attributes.items()}}]})
+        ids = payload.get("messageIds", [])
+        return ids[0] if ids else ""
+
+    def pull(self, timeout: Optional[float] = None) -> Optional[Message]:
+        import base64
+        payload = self._call(
+            "POST",
+            f"projects/{self.project}/subscriptions/{self.subscription}:pull",
+            {"maxMessages": 1})
+        received = payload.get("receivedMessages", [])
+        if not received:
+            return None
+        rm = received[0]
+        m = rm.get("message", {})
+        ack_id = rm.get("ackId", "")
+        return Message(
+            data=base64.b64decode(m.get("data", "") or ""),
+            attributes=m.get("attributes", {}) or {},
+            message_id=m.get("messageId", ack_id),
+            _on_ack=lambda _mid, a=ack_id: self._ack(a))
+
+    def _ack(self, ack_id: str) -> None:
+        self._call(
+            "POST",
+            f"projects/{self.project}/subscriptions/{self.subscription}"
+            ":acknowledge",
+            {"ackIds": [ack_id]})
+
+    # --- pubsub_util.py parity (idempotent creates) ------------------------
+    def create_topic_if_not_exists(self) -> None:
+        self._call("PUT", f"projects/{self.project}/topics/{self.topic}", {})
+
+    def create_subscription_if_not_exists(self) -> None:
+        self._call(
+            "PUT",
+            f"projects/{self.project}/subscriptions/{self.subscription}",
+            {"topic": f"projects/{self.project}/topics/{self.topic}"})
+
+
+def queue_from_env(transport: Optional[Callable] = None) -> BaseQueue:
     """Reference env contract (worker.py:68-86): PROJECT + ISSUE_EVENT_TOPIC
-    + ISSUE_EVENT_SUBSCRIPTION select Pub/Sub; otherwise a LocalQueue
-    (optionally spooled at CI_QUEUE_SPOOL)."""
+    + ISSUE_EVENT_SUBSCRIPTION select the Pub/Sub REST adapter; otherwise
+    a LocalQueue (optionally spooled at CI_QUEUE_SPOOL)."""
     import os
-    if os.environ.get("PROJECT") and os.environ.get("ISSUE_EVENT_TOPIC"):
-        try:
-            from google.cloud import pubsub  # type: ignore # noqa
-        except ImportError as e:
-            raise RuntimeError(
-                "PROJECT/ISSUE_EVENT_TOPIC set but google-cloud-pubsub is not "
-                "installed in this image; use the LocalQueue spool instead") from e
-        raise RuntimeError("Pub/Sub adapter requires network; unavailable here")
+    project = os.environ.get("PROJECT")
+    topic = os.environ.get("ISSUE_EVENT_TOPIC")
+    if project and topic:
+        q = PubSubRestQueue(
+            project, topic,
+            os.environ.get("ISSUE_EVENT_SUBSCRIPTION", f"{topic}-sub"),
+            transport=transport)
+        q.create_topic_if_not_exists()
+        q.create_subscription_if_not_exists()
+        return q
     return LocalQueue(spool_path=os.environ.get("CI_QUEUE_SPOOL"))

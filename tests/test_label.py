@@ -447,3 +447,98 @@ def test_shipped_universal_artifact_loads_and_separates():
     assert "kind/bug" in bug and "kind/feature" not in bug
     assert "kind/feature" in feat
     assert "kind/question" in q
+
+
+# ---- Pub/Sub REST adapter contract tests (recorded wire fixtures,
+# the python analogue of the reference's RoundTripper tests:
+# go/cmd/automl/pkg/client/client_test.go:18-31) -----------------------------
+
+class _RecordedTransport:
+    """Transport double: asserts request shapes, replays canned wire
+    responses, records every call."""
+
+    def __init__(self, fixtures):
+        self.fixtures = list(fixtures)
+        self.calls = []
+
+    def __call__(self, method, url, body, headers):
+        self.calls.append((method, url, body))
+        for match, (status, payload) in self.fixtures:
+            if match in url:
+                return status, payload
+        raise AssertionError(f"unexpected wire call {method} {url}")
+
+
+def test_pubsub_rest_pull_ack_contract():
+    import base64
+    from code_intelligence_amd.label.queueing import PubSubRestQueue
+    wire_msg = {"receivedMessages": [{
+        "ackId": "ACK123",
+        "message": {"data": base64.b64encode(b"payload").decode(),
+                    "attributes": {"repo_owner": "kubeflow",
+                                   "repo_name": "code-intelligence",
+                                   "issue_num": "42"},
+                    "messageId": "m-1"}}]}
+    tr = _RecordedTransport([
+        (":pull", (200, wire_msg)),
+        (":acknowledge", (200, {})),
+    ])
+    q = PubSubRestQueue("proj", "events", "bot-sub", transport=tr,
+                        token_provider=lambda: "tok")
+    msg = q.pull()
+    assert msg.data == b"payload"
+    assert msg.attributes["issue_num"] == "42"
+    msg.ack()
+    methods = [(m, u.split("/v1/")[1]) for m, u, _ in tr.calls]
+    assert methods[0] == ("POST", "projects/proj/subscriptions/bot-sub:pull")
+    assert methods[1] == ("POST",
+                          "projects/proj/subscriptions/bot-sub:acknowledge")
+    assert tr.calls[1][2] == {"ackIds": ["ACK123"]}
+    assert tr.calls[0][2] == {"maxMessages": 1}
+
+
+def test_pubsub_rest_publish_and_empty_pull():
+    import base64
+    from code_intelligence_amd.label.queueing import PubSubRestQueue
+    tr = _RecordedTransport([
+        (":publish", (200, {"messageIds": ["555"]})),
+        (":pull", (200, {})),
+    ])
+    q = PubSubRestQueue("proj", "events", "bot-sub", transport=tr)
+    mid = q.publish(b"hello", installation_id=7)
+    assert mid == "555"
+    body = tr.calls[0][2]["messages"][0]
+    assert base64.b64decode(body["data"]) == b"hello"
+    assert body["attributes"] == {"installation_id": "7"}
+    assert q.pull() is None  # empty pull -> no message, no crash
+
+
+def test_pubsub_rest_idempotent_creates_and_errors():
+    import pytest as _pytest
+    from code_intelligence_amd.label.queueing import PubSubRestQueue
+    tr = _RecordedTransport([
+        ("/topics/events", (409, {"error": {"status": "ALREADY_EXISTS"}})),
+        ("/subscriptions/bot-sub", (200, {})),
+    ])
+    q = PubSubRestQueue("proj", "events", "bot-sub", transport=tr)
+    q.create_topic_if_not_exists()       # 409 tolerated (idempotent)
+    q.create_subscription_if_not_exists()
+    assert tr.calls[1][2]["topic"] == "projects/proj/topics/events"
+    tr2 = _RecordedTransport([(":pull", (403, {"error": "denied"}))])
+    q2 = PubSubRestQueue("proj", "events", "bot-sub", transport=tr2)
+    with _pytest.raises(RuntimeError, match="403"):
+        q2.pull()
+
+
+def test_queue_from_env_selects_rest_adapter(monkeypatch):
+    from code_intelligence_amd.label import queueing
+    monkeypatch.setenv("PROJECT", "proj")
+    monkeypatch.setenv("ISSUE_EVENT_TOPIC", "events")
+    monkeypatch.setenv("ISSUE_EVENT_SUBSCRIPTION", "bot-sub")
+    tr = _RecordedTransport([
+        ("/topics/events", (200, {})),
+        ("/subscriptions/bot-sub", (409, {})),
+    ])
+    q = queueing.queue_from_env(transport=tr)
+    assert isinstance(q, queueing.PubSubRestQueue)
+    assert len(tr.calls) == 2  # idempotent create calls on construction
