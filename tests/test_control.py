@@ -162,3 +162,36 @@ def test_k8s_manifests_parse_and_mirror_reference_topology():
             "MODEL_CONFIG", "ISSUE_EMBEDDING_SERVICE"} <= env
     assert ("Service", "issue-embedding-server") in kinds
     assert ("Deployment", "modelsync") in kinds
+
+
+def test_pipeline_runner_two_step_contract(tmp_path):
+    """KFP-equivalent pipeline artifact: ordered steps, param templating,
+    fail-fast, JSON run record (reference Training_Pipeline.ipynb's 2-step
+    scrape -> train)."""
+    import json
+    from code_intelligence_amd.control.pipeline import (PipelineRunner,
+                                                        load_pipeline)
+    spec = load_pipeline("deploy/pipelines/scrape_train_pipeline.yaml")
+    assert [s.name for s in spec.steps] == ["scrape-issue-embeddings",
+                                            "train-repo-mlp"]
+    calls = []
+
+    def fake(cmd):
+        calls.append(cmd)
+        return 0
+
+    rec = PipelineRunner(spec, run_dir=tmp_path, runner=fake).run(
+        org="acme", repo="widgets")
+    assert rec["status"] == "Succeeded" and len(calls) == 2
+    assert "--org" in calls[0] and "acme" in calls[0]
+    assert "widgets" in calls[1]
+    saved = json.loads(next(tmp_path.glob("*.json")).read_text())
+    assert saved["steps"][1]["status"] == "Succeeded"
+
+    # fail-fast: step 1 failure stops the run before step 2
+    def fail_first(cmd):
+        return 3 if "embed_repo" in " ".join(cmd) else 0
+
+    rec2 = PipelineRunner(spec, runner=fail_first).run()
+    assert rec2["status"] == "Failed"
+    assert len(rec2["steps"]) == 1 and rec2["steps"][0]["returncode"] == 3
