@@ -17,6 +17,11 @@ void quantize_e4m3(at::Tensor src, at::Tensor dst, at::Tensor scale);
 void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
                            at::Tensor cs, at::Tensor gates);
+long lstm_seq_forward_gemv_persistent(at::Tensor xp, at::Tensor bias,
+                                      at::Tensor h0, at::Tensor c0,
+                                      at::Tensor w_hh, at::Tensor hs,
+                                      at::Tensor cs, at::Tensor gates,
+                                      at::Tensor ws);
 void lstm_seq_forward_gemv_fp8(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                                at::Tensor c0, at::Tensor w8, at::Tensor wscale,
                                at::Tensor hs, at::Tensor cs, at::Tensor gates);
@@ -64,6 +69,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "one-pass e4m3 quantize: dst = e4m3(src/scale)");
   m.def("lstm_seq_forward_gemv", &ci::lstm_seq_forward_gemv,
         "LSTM sequence forward (fused GEMV+cell kernel, small batch)");
+  m.def("lstm_seq_forward_gemv_persistent",
+        &ci::lstm_seq_forward_gemv_persistent,
+        "whole-sequence persistent GEMV+cell kernel (grid barrier)");
   m.def("lstm_seq_forward_gemv_fp8", &ci::lstm_seq_forward_gemv_fp8,
         "LSTM sequence forward (fp8-weight GEMV+cell kernel)");
   m.def("lstm_seq_backward", &ci::lstm_seq_backward,

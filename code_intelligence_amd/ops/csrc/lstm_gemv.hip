@@ -88,6 +88,129 @@ __global__ __launch_bounds__(256) void lstm_cell_gemv(
   }
 }
 
+// ---- persistent whole-sequence variant (CI_SERVE_PERSISTENT) ------------
+// One launch per LAYER instead of one per timestep: the grid stays
+// resident and a software grid barrier separates timesteps (NOTES r1
+// item 2: the cross-timestep persistent grid aiming at the ~3 ms
+// weight-stream floor; ~2x T x n_layers launch overheads removed).
+// Safety: the spin has a hard cap — on overflow the kernel sets a fail
+// flag and EXITS instead of hanging the GPU; the host falls back to the
+// per-step path when the flag is set. Grid size = occupancy-derived
+// resident capacity (all blocks must be co-resident for the barrier).
+
+__device__ __forceinline__ bool grid_sync_capped(unsigned int* cnt,
+                                                 unsigned int* gen,
+                                                 unsigned int nb) {
+  __syncthreads();
+  __shared__ int ok_s;
+  if (threadIdx.x == 0) {
+    ok_s = 1;
+    __threadfence();                       // publish this block's writes
+    const unsigned int g = atomicAdd(gen, 0u);   // read generation FIRST
+    if (atomicAdd(cnt, 1u) == nb - 1) {
+      atomicExch(cnt, 0u);
+      __threadfence();
+      atomicAdd(gen, 1u);                  // release the cohort
+    } else {
+      long spins = 0;
+      while (atomicAdd(gen, 0u) == g) {
+        __builtin_amdgcn_s_sleep(32);
+        if (++spins > (1 << 24)) { ok_s = 0; break; }  // bail, don't hang
+      }
+    }
+    __threadfence();                       // acquire: invalidate caches
+  }
+  __syncthreads();
+  return ok_s != 0;
+}
+
+__global__ __launch_bounds__(256) void lstm_seq_gemv_persistent(
+    const __hip_bfloat16* __restrict__ h0, long h_rs,
+    const __hip_bfloat16* __restrict__ w_hh,
+    const __hip_bfloat16* __restrict__ xp,
+    const float* __restrict__ bias,
+    const float* __restrict__ c0, long cp_rs,
+    __hip_bfloat16* __restrict__ hs,
+    float* __restrict__ cs,
+    __hip_bfloat16* __restrict__ gates_out,
+    unsigned int* __restrict__ barrier_ws,   // [cnt, gen]
+    int* __restrict__ fail_flag,
+    int B, int H, int T) {
+  constexpr int BMAX = 8;
+  const unsigned int NB = gridDim.x;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int Hv = H / 8 * 8;
+  __shared__ float dots[BMAX][4];
+  for (int t = 0; t < T; ++t) {
+    const __hip_bfloat16* hp = (t == 0) ? h0 : hs + (long)(t - 1) * B * H;
+    const float* cp = (t == 0) ? c0 : cs + (long)(t - 1) * B * H;
+    const long h_prev_rs = (t == 0) ? h_rs : (long)H;
+    const long c_prev_rs = (t == 0) ? cp_rs : (long)H;
+    const __hip_bfloat16* xpt = xp + (long)t * B * 4 * H;
+    __hip_bfloat16* h_out = hs + (long)t * B * H;
+    float* c_out = cs + (long)t * B * H;
+    __hip_bfloat16* g_out = gates_out + (long)t * B * 4 * H;
+    for (int j = blockIdx.x; j < H; j += NB) {
+      const __hip_bfloat16* wrow = w_hh + (long)(wave * H + j) * H;
+      float acc[BMAX];
+      #pragma unroll
+      for (int b = 0; b < BMAX; ++b) acc[b] = 0.f;
+      for (int k = lane * 8; k < Hv; k += 64 * 8) {
+        bf16x8g wv = *reinterpret_cast<const bf16x8g*>(wrow + k);
+        #pragma unroll
+        for (int b = 0; b < BMAX; ++b) {
+          if (b >= B) break;
+          bf16x8g hv = *reinterpret_cast<const bf16x8g*>(
+              hp + (long)b * h_prev_rs + k);
+          #pragma unroll
+          for (int e = 0; e < 8; ++e)
+            acc[b] += (float)wv[e] * (float)hv[e];
+        }
+      }
+      for (int k = Hv + lane; k < H; k += 64) {
+        const float wv = ld(wrow + k);
+        #pragma unroll
+        for (int b = 0; b < BMAX; ++b) {
+          if (b >= B) break;
+          acc[b] += wv * ld(hp + (long)b * h_prev_rs + k);
+        }
+      }
+      #pragma unroll
+      for (int b = 0; b < BMAX; ++b) {
+        if (b >= B) break;
+        float a = acc[b];
+        #pragma unroll
+        for (int off = 32; off > 0; off >>= 1)
+          a += __shfl_down(a, off);
+        if (lane == 0) dots[b][wave] = a;
+      }
+      __syncthreads();
+      for (int b = threadIdx.x; b < B; b += blockDim.x) {
+        const long xo = (long)b * 4 * H + j;
+        const float gi = sigmoidf_(dots[b][0] + ld(xpt + xo) + bias[j]);
+        const float gf = sigmoidf_(dots[b][1] + ld(xpt + xo + H) + bias[j + H]);
+        const float gg = tanhf(dots[b][2] + ld(xpt + xo + 2 * H) + bias[j + 2 * H]);
+        const float go = sigmoidf_(dots[b][3] + ld(xpt + xo + 3 * H) + bias[j + 3 * H]);
+        const float c = gf * cp[(long)b * c_prev_rs + j] + gi * gg;
+        const float h = go * tanhf(c);
+        st(h_out + (long)b * H + j, h);
+        c_out[(long)b * H + j] = c;
+        const long g0 = (long)b * 4 * H + j;
+        st(g_out + g0, gi);
+        st(g_out + g0 + H, gf);
+        st(g_out + g0 + 2 * H, gg);
+        st(g_out + g0 + 3 * H, go);
+      }
+      __syncthreads();  // dots reused by the next j of this block
+    }
+    if (!grid_sync_capped(barrier_ws, barrier_ws + 1, NB)) {
+      if (threadIdx.x == 0) atomicExch(fail_flag, 1);
+      return;
+    }
+  }
+}
+
 // whole-sequence driver over time-major (T,B,·) saves, one launch/step.
 void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
                            at::Tensor c0, at::Tensor w_hh, at::Tensor hs,
@@ -118,6 +241,45 @@ void lstm_seq_forward_gemv(at::Tensor xp, at::Tensor bias, at::Tensor h0,
   }
 }
 
+
+// persistent driver: ONE launch for the whole sequence. Returns the grid
+// size used, or 0 when a co-resident grid is not available (caller falls
+// back to the per-step driver). ws = int32[4] zeroed workspace
+// [cnt, gen, fail, pad]; caller checks ws[2] after the stream syncs.
+long lstm_seq_forward_gemv_persistent(at::Tensor xp, at::Tensor bias,
+                                      at::Tensor h0, at::Tensor c0,
+                                      at::Tensor w_hh, at::Tensor hs,
+                                      at::Tensor cs, at::Tensor gates,
+                                      at::Tensor ws) {
+  CI_CHECK_CUDA(xp); CI_CHECK_CONTIG(xp); CI_CHECK_CONTIG(hs);
+  CI_CHECK_CONTIG(cs); CI_CHECK_CONTIG(gates); CI_CHECK_CONTIG(ws);
+  TORCH_CHECK(xp.scalar_type() == at::ScalarType::BFloat16);
+  TORCH_CHECK(ws.numel() >= 4 && ws.scalar_type() == at::ScalarType::Int);
+  const int T = xp.size(0), B = xp.size(1);
+  TORCH_CHECK(B <= 8, "gemv persistent kernel is for B <= 8");
+  const int H = w_hh.size(1);
+  int per_cu = 0;
+  if (hipOccupancyMaxActiveBlocksPerMultiprocessor(
+          &per_cu, reinterpret_cast<const void*>(&lstm_seq_gemv_persistent),
+          256, 0) != hipSuccess || per_cu < 1) {
+    return 0;
+  }
+  const int cus = at::cuda::getCurrentDeviceProperties()->multiProcessorCount;
+  const long nb = std::min<long>((long)per_cu * cus, H);
+  auto h0c = h0.contiguous();
+  hipLaunchKernelGGL(lstm_seq_gemv_persistent, dim3(nb), dim3(256), 0,
+      stream(),
+      reinterpret_cast<const __hip_bfloat16*>(h0c.data_ptr()), (long)H,
+      reinterpret_cast<const __hip_bfloat16*>(w_hh.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(xp.data_ptr()),
+      bias.data_ptr<float>(), c0.data_ptr<float>(), (long)H,
+      reinterpret_cast<__hip_bfloat16*>(hs.data_ptr()),
+      cs.data_ptr<float>(),
+      reinterpret_cast<__hip_bfloat16*>(gates.data_ptr()),
+      reinterpret_cast<unsigned int*>(ws.data_ptr()),
+      reinterpret_cast<int*>(ws.data_ptr()) + 2, B, H, T);
+  return nb;
+}
 
 // ---- fp8 (OCP e4m3) weight variant --------------------------------------
 // Serving is bound by streaming W_hh (46 MB/layer-step at the deployed

@@ -166,6 +166,21 @@ class _FusedLSTMFunction(torch.autograd.Function):
                 lib.lstm_seq_forward_gemv_fp8(xp, bias, h0,
                                               c0.to(torch.float32), w8,
                                               wscale, hs, cs, gates)
+            elif os.environ.get("CI_SERVE_PERSISTENT", "0") == "1":
+                # whole-sequence persistent grid: one launch per layer,
+                # software grid barrier between timesteps. The kernel
+                # bails (fail flag) instead of hanging if the grid is
+                # not co-resident; fall back per-step then.
+                ws = torch.zeros(4, dtype=torch.int32, device=x.device)
+                c32 = c0.to(torch.float32)
+                nb = lib.lstm_seq_forward_gemv_persistent(
+                    xp, bias, h0, c32, w_hh, hs, cs, gates, ws)
+                if nb == 0 or int(ws[2].item()) != 0:
+                    import warnings
+                    warnings.warn("persistent GEMV unavailable "
+                                  f"(nb={nb}); per-step fallback")
+                    lib.lstm_seq_forward_gemv(xp, bias, h0, c32,
+                                              w_hh, hs, cs, gates)
             else:
                 lib.lstm_seq_forward_gemv(xp, bias, h0, c0.to(torch.float32),
                                           w_hh, hs, cs, gates)

@@ -636,3 +636,61 @@ def test_quantize_e4m3_kernel():
     lib.quantize_e4m3(x, q, s)
     ref = ((x.float() / s).clamp(-448, 448)).to(torch.float8_e4m3fn)
     assert torch.equal(q.view(torch.uint8), ref.view(torch.uint8))
+
+
+# ---- persistent whole-sequence GEMV (CI_SERVE_PERSISTENT, round 2) --------
+
+@pytest.mark.timeout(300)
+def test_persistent_gemv_matches_per_step():
+    from code_intelligence_amd.ops import extension as ext
+    lib = ext.require()
+    torch.manual_seed(33)
+    for B, T, H in [(1, 40, 2400), (4, 17, 96), (2, 3, 50)]:
+        xp = torch.randn(T, B, 4 * H, device=DEV, dtype=torch.bfloat16) * 0.3
+        bias = torch.randn(4 * H, device=DEV, dtype=torch.float32) * 0.05
+        h0 = torch.randn(B, H, device=DEV, dtype=torch.bfloat16) * 0.3
+        c0 = torch.randn(B, H, device=DEV, dtype=torch.float32) * 0.3
+        w = torch.randn(4 * H, H, device=DEV, dtype=torch.bfloat16) * 0.05
+
+        def bufs():
+            return (torch.empty(T, B, H, device=DEV, dtype=torch.bfloat16),
+                    torch.empty(T, B, H, device=DEV, dtype=torch.float32),
+                    torch.empty(T, B, 4 * H, device=DEV,
+                                dtype=torch.bfloat16))
+
+        hs1, cs1, g1 = bufs()
+        lib.lstm_seq_forward_gemv(xp, bias, h0, c0, w, hs1, cs1, g1)
+        hs2, cs2, g2 = bufs()
+        ws = torch.zeros(4, dtype=torch.int32, device=DEV)
+        nb = lib.lstm_seq_forward_gemv_persistent(xp, bias, h0, c0, w,
+                                                  hs2, cs2, g2, ws)
+        torch.cuda.synchronize()
+        assert nb > 0, "persistent grid unavailable"
+        assert int(ws[2]) == 0, "grid barrier bailed"
+        assert torch.equal(hs1, hs2), (B, T, H,
+                                       (hs1.float() - hs2.float()).abs().max())
+        assert torch.allclose(cs1, cs2, atol=1e-5)
+        assert torch.equal(g1, g2)
+
+
+def test_persistent_gemv_serve_wrapper_path():
+    """InferenceWrapper single-request path under CI_SERVE_PERSISTENT=1
+    equals the default path's embedding."""
+    import numpy as np
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.engine.inference import InferenceWrapper
+    from code_intelligence_amd.text.tokenizer import Vocab, defaults_specials
+    torch.manual_seed(34)
+    words = [f"w{i}" for i in range(3000)]
+    v = Vocab(defaults_specials + words)
+    model = AWDLSTM(vocab_sz=len(v), emb_sz=64, n_hid=96, n_layers=2)
+    w = InferenceWrapper(encoder=model.encoder, vocab=v)
+    text = "xxfld 1 crash w7 w8 w9 xxfld 2 " + " ".join(
+        f"w{i % 2900}" for i in range(60))
+    a = w.get_pooled_features(text).numpy()
+    os.environ["CI_SERVE_PERSISTENT"] = "1"
+    try:
+        b = w.get_pooled_features(text).numpy()
+    finally:
+        os.environ.pop("CI_SERVE_PERSISTENT", None)
+    np.testing.assert_allclose(a, b, atol=1e-3)
