@@ -115,7 +115,7 @@ __device__ __forceinline__ bool grid_sync_capped(unsigned int* cnt,
       long spins = 0;
       while (atomicAdd(gen, 0u) == g) {
         __builtin_amdgcn_s_sleep(32);
-        if (++spins > (1 << 24)) { ok_s = 0; break; }  // bail, don't hang
+        if (++spins > (1 << 20)) { ok_s = 0; break; }  // bail, don't hang
       }
     }
     __threadfence();                       // acquire: invalidate caches
@@ -265,7 +265,10 @@ long lstm_seq_forward_gemv_persistent(at::Tensor xp, at::Tensor bias,
     return 0;
   }
   const int cus = at::cuda::getCurrentDeviceProperties()->multiProcessorCount;
-  const long nb = std::min<long>((long)per_cu * cus, H);
+  long nb = std::min<long>((long)per_cu * cus, H);
+  if (const char* env = getenv("CI_PERS_NB")) {   // residency experiments
+    nb = std::min<long>(std::max(1L, atol(env)), H);
+  }
   auto h0c = h0.contiguous();
   hipLaunchKernelGGL(lstm_seq_gemv_persistent, dim3(nb), dim3(256), 0,
       stream(),

@@ -166,7 +166,8 @@ class _FusedLSTMFunction(torch.autograd.Function):
                 lib.lstm_seq_forward_gemv_fp8(xp, bias, h0,
                                               c0.to(torch.float32), w8,
                                               wscale, hs, cs, gates)
-            elif os.environ.get("CI_SERVE_PERSISTENT", "0") == "1":
+            elif os.environ.get("CI_SERVE_PERSISTENT", "0") == "1" \
+                    and not getattr(_FusedLSTMFunction, "_pers_failed", False):
                 # whole-sequence persistent grid: one launch per layer,
                 # software grid barrier between timesteps. The kernel
                 # bails (fail flag) instead of hanging if the grid is
@@ -178,7 +179,8 @@ class _FusedLSTMFunction(torch.autograd.Function):
                 if nb == 0 or int(ws[2].item()) != 0:
                     import warnings
                     warnings.warn("persistent GEMV unavailable "
-                                  f"(nb={nb}); per-step fallback")
+                                  f"(nb={nb}); per-step fallback (sticky)")
+                    _FusedLSTMFunction._pers_failed = True  # stop re-bailing
                     lib.lstm_seq_forward_gemv(xp, bias, h0, c32,
                                               w_hh, hs, cs, gates)
             else:
