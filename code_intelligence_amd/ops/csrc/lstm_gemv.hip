@@ -97,6 +97,15 @@ __global__ __launch_bounds__(256) void lstm_cell_gemv(
 // flag and EXITS instead of hanging the GPU; the host falls back to the
 // per-step path when the flag is set. Grid size = occupancy-derived
 // resident capacity (all blocks must be co-resident for the barrier).
+//
+// MEASURED NEGATIVE (kept as a documented experiment, opt-in only):
+// at the deployed shape the barrier itself costs ~50 us/step — the
+// generation counter bounces across all 8 XCDs' L2s — so the best
+// persistent config (nb=256, 15.4 ms at T=300) is ~4.7x SLOWER than
+// per-step launches (3.3 ms). gpurun_out/r2m_pers.log /
+// profiles/BENCH_HISTORY.md. Also: hipOccupancyMaxActiveBlocks
+// overestimates by 1 block/CU here (claimed 7, 6 resident), hence the
+// safety margin below.
 
 __device__ __forceinline__ bool grid_sync_capped(unsigned int* cnt,
                                                  unsigned int* gen,
@@ -265,6 +274,7 @@ long lstm_seq_forward_gemv_persistent(at::Tensor xp, at::Tensor bias,
     return 0;
   }
   const int cus = at::cuda::getCurrentDeviceProperties()->multiProcessorCount;
+  per_cu = std::max(1, per_cu - 1);  // measured: API claims 7/CU, 6 resident
   long nb = std::min<long>((long)per_cu * cus, H);
   if (const char* env = getenv("CI_PERS_NB")) {   // residency experiments
     nb = std::min<long>(std::max(1L, atol(env)), H);
