@@ -190,7 +190,13 @@ static void launch_fwd_step(const at::Tensor& xp, const at::Tensor& bias,
   // TIME-MAJOR layout: xp/hs/cs/gates are (T, B, ·) contiguous, so slice t
   // is a contiguous (B, ·) block — hipBLASLt sees contiguous operands and
   // the cell kernel gets unit row strides.
-  const int threads = 256;
+  // 128-thread blocks double the workgroup count (the deployed shape
+  // yields only 600 WGs at 256 threads = 2.3/CU and the kernel is
+  // latency-bound at 43% of HBM peak); CI_CELL_THREADS overrides.
+  static const int threads = [] {
+    const char* e = getenv("CI_CELL_THREADS");
+    return e ? atoi(e) : 128;
+  }();
   constexpr int VEC = 16 / sizeof(ST);
   const int Hv = H / VEC;
   const ST* xpp = reinterpret_cast<const ST*>(xp.data_ptr()) + (long)t * B * 4 * H;
@@ -298,7 +304,10 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
   // (scripts/gemm_probe.py on MI355X)
   auto w_hh_tc = w_hh.t().contiguous();
   auto w_hh_nt = w_hh_tc.t();
-  const int threads = 256;
+  static const int threads = [] {
+    const char* e = getenv("CI_CELL_THREADS");
+    return e ? atoi(e) : 128;
+  }();
   CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
     constexpr int VEC = 16 / sizeof(scalar_t);
     const bool vec_ok = (H % VEC) == 0;
