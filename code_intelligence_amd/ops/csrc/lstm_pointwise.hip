@@ -190,12 +190,13 @@ static void launch_fwd_step(const at::Tensor& xp, const at::Tensor& bias,
   // TIME-MAJOR layout: xp/hs/cs/gates are (T, B, ·) contiguous, so slice t
   // is a contiguous (B, ·) block — hipBLASLt sees contiguous operands and
   // the cell kernel gets unit row strides.
-  // 128-thread blocks double the workgroup count (the deployed shape
+  // 64-thread blocks quadruple the workgroup count (the deployed shape
   // yields only 600 WGs at 256 threads = 2.3/CU and the kernel is
-  // latency-bound at 43% of HBM peak); CI_CELL_THREADS overrides.
+  // latency-bound; measured 432.3 vs 435.2 ms/step e2e at 64 vs 256) —
+  // CI_CELL_THREADS overrides.
   static const int threads = [] {
     const char* e = getenv("CI_CELL_THREADS");
-    return e ? atoi(e) : 128;
+    return e ? atoi(e) : 64;
   }();
   constexpr int VEC = 16 / sizeof(ST);
   const int Hv = H / VEC;
@@ -306,7 +307,7 @@ void lstm_seq_backward(at::Tensor dhs, at::Tensor dhT, at::Tensor dcT,
   auto w_hh_nt = w_hh_tc.t();
   static const int threads = [] {
     const char* e = getenv("CI_CELL_THREADS");
-    return e ? atoi(e) : 128;
+    return e ? atoi(e) : 64;
   }();
   CI_DISPATCH_FB(dhs.scalar_type(), "lstm_seq_backward", [&] {
     constexpr int VEC = 16 / sizeof(scalar_t);
