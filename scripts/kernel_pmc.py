@@ -90,3 +90,42 @@ for _ in range(3):
 
 torch.cuda.synchronize()
 print("pmc workload done (incl qrnn)")
+
+# ---- round-2 kernels -----------------------------------------------------
+# dual dlogits epilogue (fp8 CE backward): bf16 in-place + e4m3 scratch
+scratch8 = torch.empty(N, V, device=dev, dtype=torch.float8_e4m3fn)
+for _ in range(3):
+    lib.ce_dlogits_dual(logits, tgt, b32, lse, scale, scratch8, 448.0)
+
+# fused AR/TAR fwd/bwd at the bench shape
+out_a = torch.randn(512, 512, 800, device=dev, dtype=dt)
+r_tm = torch.randn(512, 512, 800, device=dev, dtype=dt)  # (T,B,H)
+d32 = torch.full((1,), 1.0, device=dev, dtype=torch.float32)
+for _ in range(3):
+    lib.artar_forward(out_a, r_tm)
+    lib.artar_backward(out_a, r_tm, d32, 1e-8, 1e-8)
+
+# K1 gather/scatter at the deployed shape
+Vemb, E = 60000, 800
+wemb = torch.randn(Vemb, E, device=dev, dtype=dt)
+ids = torch.randint(0, Vemb, (512, 512), device=dev)
+rowmask = (torch.rand(Vemb, device=dev) > 0.02).float()
+gout = torch.randn(512, 512, E, device=dev, dtype=dt)
+for _ in range(3):
+    lib.emb_gather(wemb, ids, rowmask)
+    lib.emb_scatter(gout, ids, rowmask, Vemb, 1)
+
+# K3 seeded dropconnect apply/grad on the 4Hx H weight
+for _ in range(3):
+    wm = lib.dropconnect_apply(w, 12345, 0.2)
+    lib.dropconnect_grad_(wm, 12345, 0.2)
+
+# one-pass quantize at the projection shape
+src = torch.randn(262144, 2400, device=dev, dtype=dt)
+q8 = torch.empty(262144, 2400, device=dev, dtype=torch.float8_e4m3fn)
+sc1 = torch.full((), 0.01, device=dev, dtype=torch.float32)
+for _ in range(3):
+    lib.quantize_e4m3(src, q8, sc1)
+
+torch.cuda.synchronize()
+print("pmc workload done (incl round-2 kernels)")
