@@ -121,10 +121,11 @@ class _FusedCEFp8Function(torch.autograd.Function):
         w8_t = w8.t()  # (H, V) column-major view for mat2
         w_t = weight.t()
         logits_full = None
-        try:
-            logits_full = torch.empty(N, V, dtype=h.dtype, device=dev)
-        except torch.cuda.OutOfMemoryError:
-            pass
+        if os.environ.get("CI_CE_SAVE_LOGITS", "1") != "0":
+            try:
+                logits_full = torch.empty(N, V, dtype=h.dtype, device=dev)
+            except torch.cuda.OutOfMemoryError:
+                pass
         scratch = None if logits_full is not None else \
             torch.empty(min(C, N), V, dtype=h.dtype, device=dev)
         for s in range(0, N, C):

@@ -193,7 +193,11 @@ class _FusedLSTMFunction(torch.autograd.Function):
             else:
                 lib.lstm_seq_forward_gemv(xp, bias, h0, c0.to(torch.float32),
                                           w_hh, hs, cs, gates)
-        elif mode == "fused" and dt == torch.bfloat16 and H % 8 == 0:
+        elif dt == torch.bfloat16 and H % 8 == 0 and (
+                mode == "fused"
+                or (mode == "auto" and H >= 2048 and B >= 128)):
+            # "auto": fused MFMA cell only where its grid fills the chip
+            # (H=2400 layers -> 600 blocks); small layers stay on lib
             lib.lstm_seq_forward_fused(xp, bias, h0, c0.to(torch.float32), w_hh,
                                        hs, cs, gates)
         elif fp8 and mode == "lib":
