@@ -110,3 +110,19 @@ def test_converted_artifacts_golden_vectors(fake_pkl, tmp_path):
         b = w_ref.get_pooled_features(t_ref).numpy()
         assert a.shape == (1, 3 * E)
         np.testing.assert_allclose(a, b, atol=1e-6)
+
+
+def test_convert_errors_without_encoder_or_vocab(tmp_path):
+    """Malformed pickles raise clear ValueErrors instead of writing
+    partial artifacts."""
+    import pytest as _pytest
+    from scripts.convert_fastai_pkl import convert
+    torch.save({"just": "a dict", "tensor": torch.zeros(3)},
+               tmp_path / "noenc.pkl")
+    with _pytest.raises(ValueError, match="no AWD_LSTM encoder"):
+        convert(tmp_path / "noenc.pkl", tmp_path / "out1")
+
+    enc_only = AWDLSTM(vocab_sz=64, emb_sz=16, n_hid=24, n_layers=2).encoder
+    torch.save({"model": enc_only}, tmp_path / "novocab.pkl")
+    with _pytest.raises(ValueError, match="no vocab"):
+        convert(tmp_path / "novocab.pkl", tmp_path / "out2")
