@@ -204,6 +204,8 @@ def main():
         del trainer, model
         if on_gpu:
             torch.cuda.empty_cache()
+            time.sleep(3)  # let clocks recover from the training burst so
+            # the serve metric measures serving, not DVFS aftermath
         serve = serve_bench(args, quick=True)
 
     if rank == 0:
