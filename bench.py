@@ -139,6 +139,14 @@ def main():
         return
     rank, world = init_distributed()
     on_gpu = torch.cuda.is_available()
+    serve = None
+    if rank == 0 and world == 1 and not args.no_serve:
+        # BASELINE.json's metric names BOTH halves ("LM tokens/sec ... and
+        # issue-embeddings/sec served"): measure serving FIRST, before the
+        # sustained training burst drags clocks down.
+        serve = serve_bench(args, quick=True)
+        if on_gpu:
+            torch.cuda.empty_cache()
     if not on_gpu:
         # CPU fallback: plumbing config (BASELINE.json config 1) so the
         # script stays runnable off-GPU; the official metric is GPU-only.
@@ -194,19 +202,6 @@ def main():
     n_gpus = world if on_gpu else args.gpus
     tokens = args.bs * args.seq * args.steps * world
     value = tokens / elapsed
-
-    serve = None
-    if rank == 0 and world == 1 and not args.no_serve:
-        # BASELINE.json's metric names BOTH halves ("LM tokens/sec ... and
-        # issue-embeddings/sec served"): attach a driver-timed serve
-        # measurement to the same JSON line. Free the
-        # training state first so the serve wrapper starts clean.
-        del trainer, model
-        if on_gpu:
-            torch.cuda.empty_cache()
-            time.sleep(3)  # let clocks recover from the training burst so
-            # the serve metric measures serving, not DVFS aftermath
-        serve = serve_bench(args, quick=True)
 
     if rank == 0:
         print(json.dumps({
