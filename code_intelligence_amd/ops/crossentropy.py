@@ -96,6 +96,22 @@ def _fp8r_enabled() -> bool:
 
 _STORE = 448.0  # fixed dlogits store scale: |softmax - onehot| <= 1
 
+_ce_side = None
+
+
+def _ce_side_stream():
+    global _ce_side
+    if _ce_side is None:
+        _ce_side = torch.cuda.Stream()
+    return _ce_side
+
+
+def _ce_overlap_enabled() -> bool:
+    """CI_CE_2STREAM: run each chunk's dW/db GEMMs on a side stream so
+    they overlap the next chunk's dlogits epilogue + fp8 dh GEMM (the two
+    halves touch disjoint chunk slices). Opt-in until measured."""
+    return os.environ.get("CI_CE_2STREAM", "0") == "1"
+
 
 class _FusedCEFp8Function(torch.autograd.Function):
     """fp8-GEMM variant of the fused tied-decoder CE (see _fp8r_enabled).
@@ -177,6 +193,9 @@ class _FusedCEFp8Function(torch.autograd.Function):
         w8_cm = w8.t().contiguous().t()
         w_t = weight.t()
         bias_arg = b32 if has_bias else _empty_f32(dev)
+        overlap = _ce_overlap_enabled() and logits_full is not None
+        side = _ce_side_stream() if overlap else None
+        main = torch.cuda.current_stream() if overlap else None
         for s in range(0, N, C):
             e = min(N, s + C)
             c = e - s
@@ -194,9 +213,20 @@ class _FusedCEFp8Function(torch.autograd.Function):
                                  scale_b=sw, out_dtype=h.dtype, out=dh[s:e])
             else:
                 torch.mm(dlog, weight, out=dh[s:e])
-            dw += torch.mm(dlog.t(), h[s:e])
-            if has_bias:
-                db += dlog.sum(dim=0).to(torch.float32)
+            if overlap:
+                # dW/db of THIS chunk overlap the next chunk's epilogue +
+                # dh (disjoint slices of the resident buffer)
+                side.wait_stream(main)
+                with torch.cuda.stream(side):
+                    dw += torch.mm(dlog.t(), h[s:e])
+                    if has_bias:
+                        db += dlog.sum(dim=0).to(torch.float32)
+            else:
+                dw += torch.mm(dlog.t(), h[s:e])
+                if has_bias:
+                    db += dlog.sum(dim=0).to(torch.float32)
+        if overlap:
+            main.wait_stream(side)
         return (dh, dw.to(weight.dtype),
                 db.to(ctx.bias_dtype) if has_bias else None, None)
 
