@@ -226,6 +226,8 @@ def main():
             },
             **({"serve": serve} if serve is not None else {}),
         }))
+    if world > 1 and torch.distributed.is_initialized():
+        torch.distributed.destroy_process_group()
 
 
 if __name__ == "__main__":

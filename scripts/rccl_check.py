@@ -39,3 +39,4 @@ for p in m.parameters():
 torch.cuda.synchronize()
 if rank == 0:
     print(f"rccl ok: world={world} backend=nccl loss={float(loss):.3f}")
+torch.distributed.destroy_process_group()
