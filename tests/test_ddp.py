@@ -307,3 +307,52 @@ def test_ddp_world4_training_matches_single_process(tmp_path):
     for n, p_ in m.named_parameters():
         assert torch.allclose(got["params"][n], p_.detach(),
                               atol=1e-5, rtol=1e-4), n
+
+
+def _worker_partial(rank, out_path):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(WORLD),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29551",
+                      LOCAL_RANK=str(rank))
+    dist.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        m = _build_model()
+        tr = LMTrainer(m, TrainConfig(alpha=0, beta=0), distributed=True)
+        x, y = _make_batches(seed=600 + rank)
+        # only the DECODER side of the graph gets gradients: encoder rnn
+        # params receive None grads and the bucket fallback path must
+        # zero-fill + reduce them without deadlock
+        tr.dist.prepare()
+        from code_intelligence_amd.ops.crossentropy import tied_decoder_ce
+        emb = m.encoder.encoder(x)  # embedding only, skip the rnns
+        h = emb.reshape(-1, emb.shape[-1])
+        loss = tied_decoder_ce(h, m.decoder.decoder.weight,
+                               m.decoder.decoder.bias, y.reshape(-1))
+        loss.backward()
+        tr.dist.finalize()
+        if rank == 0:
+            grads = {n: (p.grad.clone() if p.grad is not None else None)
+                     for n, p in m.named_parameters()}
+            torch.save({k: v for k, v in grads.items() if v is not None},
+                       out_path)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_ddp_partial_graph_no_deadlock(tmp_path):
+    """Params with no grads this step (untouched rnns) must not stall the
+    bucketer: finalize's zero-fill fallback reduces synchronously
+    (the transfer/fine-tune partial-graph case at scale)."""
+    ctx = mp.get_context("spawn")
+    out = str(tmp_path / "gp.pt")
+    procs = [ctx.Process(target=_worker_partial, args=(r, out))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=100)
+        assert p.exitcode == 0
+    got = torch.load(out, weights_only=True)
+    assert any("decoder" in k or "encoder.weight" in k for k in got)
+    for v in got.values():
+        assert torch.isfinite(v).all()

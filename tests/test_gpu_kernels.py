@@ -728,3 +728,22 @@ def test_ce_fp8_extreme_logits_no_overflow():
     finally:
         os.environ.pop("CI_CE_FP8R", None)
     assert abs(float(loss) - float(ref)) / max(abs(float(ref)), 1e-6) < 0.1
+
+
+def test_ce_fp8_engages_by_default():
+    """The DEFAULT environment must route the training CE through the
+    fp8 function (CI_CE_FP8R defaults on) — guards against silent
+    fallback to the bf16 path in the driver's bench."""
+    from code_intelligence_amd.ops.crossentropy import tied_decoder_ce
+    assert os.environ.get("CI_CE_FP8R") in (None, "1")
+    torch.manual_seed(51)
+    h = torch.randn(64, 32, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(512, 32, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.zeros(512, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    t = torch.randint(0, 512, (64,), device=DEV)
+    loss = tied_decoder_ce(h, w, b, t)
+    assert "FusedCEFp8" in type(loss.grad_fn).__name__, \
+        type(loss.grad_fn).__name__
