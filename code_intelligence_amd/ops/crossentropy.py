@@ -40,40 +40,6 @@ def _empty_f32(device) -> Tensor:
     return _EMPTY[key]
 
 
-def _fp8_enabled() -> bool:
-    """Opt-in fp8 (OCP e4m3) path for the CE GEMMs: MI355X runs fp8 MFMA at
-    2x the bf16 rate (measured +40% on the chunk GEMM, scripts/fp8_probe.py)
-    at ~3.6% max relative logits error from per-tensor scaling. Off by
-    default — accuracy first; enable with CI_CE_FP8=1."""
-    return os.environ.get("CI_CE_FP8", "0") == "1"
-
-
-def _to_fp8(t: Tensor):
-    """Per-tensor-scaled OCP e4m3 quantization; returns (fp8, scale_f32)."""
-    scale = (t.abs().amax().float() / 448.0).clamp_min(1e-12)
-    q = (t.float() / scale).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
-    return q, scale
-
-
-def _mm_maybe_fp8(a: Tensor, b_t: Tensor, out: Tensor,
-                  b_fp8_cache: dict | None = None, key: str = "") -> None:
-    """out = a @ b_t (b_t is a transposed view of a row-major (V,H) weight,
-    i.e. column-major — exactly the layout _scaled_mm wants for mat2)."""
-    if _fp8_enabled() and a.dtype == torch.bfloat16 and a.shape[1] % 16 == 0             and a.shape[0] % 16 == 0:
-        if b_fp8_cache is not None and key in b_fp8_cache:
-            b8, bs = b_fp8_cache[key]
-        else:
-            bq, bs = _to_fp8(b_t.t().contiguous())   # quantize row-major (V,H)
-            b8 = bq.t()                               # column-major view
-            if b_fp8_cache is not None:
-                b_fp8_cache[key] = (b8, bs)
-        a8, as_ = _to_fp8(a)
-        out.copy_(torch._scaled_mm(a8, b8, scale_a=as_, scale_b=bs,
-                                   out_dtype=out.dtype))
-        return
-    torch.mm(a, b_t, out=out)
-
-
 def _fp8r_enabled() -> bool:
     """fp8 CE GEMMs (CI_CE_FP8R): the logits GEMM runs with OCP e4m3
     INPUTS (per-tensor scales) at the fp8 MFMA rate into the bf16-resident
@@ -254,12 +220,11 @@ class _FusedCEFunction(torch.autograd.Function):
         scratch = None if logits_full is not None else \
             torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
         w_t = weight.t()
-        fp8_cache: dict = {}
         for s in range(0, N, C):
             e = min(N, s + C)
             logits = logits_full[s:e] if logits_full is not None \
                 else scratch[: e - s]
-            _mm_maybe_fp8(h[s:e], w_t, logits, fp8_cache, "w")
+            torch.mm(h[s:e], w_t, out=logits)
             lib.ce_rowstats(logits, tgt64[s:e], b32, lse[s:e], tgt_logit[s:e])
         loss = (lse - tgt_logit).mean()
         ctx.save_for_backward(h, weight, b32, tgt64, lse)
