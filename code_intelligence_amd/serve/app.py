@@ -56,6 +56,17 @@ def create_app(wrapper: InferenceWrapper | None = None,
             model_path=_resolve_model_path(model_path),
             use_graphs=os.environ.get("CI_SERVE_GRAPHS", "0") == "1")
     app.config["wrapper"] = wrapper
+    # dynamic micro-batching (CI_SERVE_BATCH_MS>0): concurrent /text
+    # callers share GPU batches through the bulk path — the reference
+    # needed 9 replicas for concurrency; one MI355X process batches it
+    batch_ms = float(os.environ.get("CI_SERVE_BATCH_MS", "0"))
+    batcher = None
+    if batch_ms > 0:
+        from .batcher import MicroBatcher
+        batcher = MicroBatcher(wrapper, window_ms=batch_ms,
+                               max_batch=int(os.environ.get(
+                                   "CI_SERVE_BATCH_MAX", "64")))
+    app.config["batcher"] = batcher
 
     @app.route("/healthz", methods=["GET"])
     def healthz():
@@ -81,7 +92,11 @@ def create_app(wrapper: InferenceWrapper | None = None,
         data = request.get_json(force=True)
         doc = wrapper.process_dict({"title": data.get("title", ""),
                                     "body": data.get("body", "")})
-        emb = wrapper.get_pooled_features(doc["text"]).numpy().astype("<f4")
+        if batcher is not None:
+            emb = batcher.embed(doc["text"]).astype("<f4")
+        else:
+            emb = wrapper.get_pooled_features(doc["text"]).numpy() \
+                .astype("<f4")
         payload = emb.tobytes()
         md5 = hashlib.md5(payload).hexdigest()
         log.debug("embedding md5=%s", md5)
@@ -122,8 +137,11 @@ def main():
     p.add_argument("--port", type=int, default=8080)
     args = p.parse_args()
     app = create_app(model_path=args.model_path)
-    # mirror the reference: debug mode is forbidden in serving (app.py:122-128)
-    app.run(host=args.host, port=args.port, debug=False, threaded=False)
+    # mirror the reference: debug mode is forbidden in serving
+    # (app.py:122-128). Single-threaded unless micro-batching is on —
+    # batching needs concurrent request threads to batch across.
+    threaded = app.config.get("batcher") is not None
+    app.run(host=args.host, port=args.port, debug=False, threaded=threaded)
 
 
 if __name__ == "__main__":

@@ -203,3 +203,79 @@ def test_wire_level_http_contract(tmp_path):
         assert r2.json()["shape"] == [2, 48]
     finally:
         srv.shutdown()
+
+
+def test_microbatcher_batches_concurrent_requests():
+    """Concurrent embed() calls share a GPU batch: 12 callers inside one
+    window must produce FEWER batches than requests, every caller gets
+    its own correct row, and errors propagate per caller."""
+    import threading
+    import numpy as np
+    from code_intelligence_amd.serve.batcher import MicroBatcher
+
+    calls = []
+
+    class _FakeWrapper:
+        def texts_to_embedding(self, texts, bs=64):
+            calls.append(list(texts))
+            # row i encodes a hash of its text so callers can verify
+            return np.stack([np.full(4, float(len(t)), dtype=np.float32)
+                             for t in texts])
+
+    mb = MicroBatcher(_FakeWrapper(), window_ms=50.0, max_batch=64)
+    try:
+        results = {}
+        errs = []
+
+        def call(i):
+            try:
+                results[i] = mb.embed("x" * (i + 1))
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        threads = [threading.Thread(target=call, args=(i,))
+                   for i in range(12)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=10)
+        assert not errs
+        assert len(results) == 12
+        for i, emb in results.items():
+            assert emb.shape == (1, 4)
+            assert float(emb[0, 0]) == i + 1  # own row, not a neighbor's
+        assert mb.batches < 12, mb.batches  # actually batched
+        assert mb.batched_requests == 12
+    finally:
+        mb.close()
+
+
+def test_microbatcher_propagates_errors():
+    from code_intelligence_amd.serve.batcher import MicroBatcher
+    import pytest as _pytest
+
+    class _Boom:
+        def texts_to_embedding(self, texts, bs=64):
+            raise RuntimeError("encode failed")
+
+    mb = MicroBatcher(_Boom(), window_ms=1.0)
+    try:
+        with _pytest.raises(RuntimeError, match="encode failed"):
+            mb.embed("hello")
+    finally:
+        mb.close()
+
+
+def test_serve_app_batched_text_route(monkeypatch, tmp_path):
+    """CI_SERVE_BATCH_MS>0: the /text route returns the same payload
+    contract through the batcher."""
+    monkeypatch.setenv("CI_SERVE_BATCH_MS", "2")
+    app = create_app(wrapper=_tiny_wrapper(tmp_path))
+    assert app.config["batcher"] is not None
+    client = app.test_client()
+    r = client.post("/text", json={"title": "hello", "body": "world"})
+    assert r.status_code == 200
+    emb = np.frombuffer(r.data, "<f4")
+    assert emb.ndim == 1 and emb.size > 0 and np.isfinite(emb).all()
+    assert "X-Embedding-MD5" in r.headers
+    app.config["batcher"].close()
