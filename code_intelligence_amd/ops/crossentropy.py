@@ -3,8 +3,8 @@
 Reference semantics: fastai LinearDecoder + FlattenedLoss(CrossEntropy)
 over a 60k vocab (train.py:70, tie_weights/out_bias).
 
-MI355X design: chunked over rows (CHUNK=32768 since r2 — measured
-+0.3-0.5% over 16384 across three boxes; ~1.05 PF for the chunk
+MI355X design: chunked over rows (CHUNK=65536 since r2 — the chunk
+ladder measured 16384 -> 32768 -> 65536 at +0.3-0.5% each; ~1.05 PF for the chunk
 GEMM — scripts/gemm_probe.py): per chunk one plain hipBLASLt GEMM fills a
 logits tile and a HIP kernel reduces it to (logsumexp, target-logit) in a
 single online pass with the decoder bias folded in (the (chunk, 60k) bias
@@ -198,7 +198,7 @@ class _FusedCEFp8Function(torch.autograd.Function):
 
 
 class _FusedCEFunction(torch.autograd.Function):
-    CHUNK = int(os.environ.get("CI_CE_CHUNK", "32768"))
+    CHUNK = int(os.environ.get("CI_CE_CHUNK", "65536"))
 
     @staticmethod
     def forward(ctx, h: Tensor, weight: Tensor, bias: Tensor, targets: Tensor):
