@@ -148,11 +148,20 @@ class _FusedCEFp8Function(torch.autograd.Function):
         C = _FusedCEFunction.CHUNK
         dev = h.device
         dh = torch.empty_like(h)
-        dw = torch.zeros_like(weight, dtype=torch.float32)
-        db = torch.zeros(V, dtype=torch.float32, device=dev) if has_bias else None
         scale = (dloss / N).to(torch.float32).reshape(1)
         sc_over_store = (scale.reshape(()) / _STORE)
         scratch8 = torch.empty(min(C, N), V, dtype=f8, device=dev)
+        # db folded into the dW GEMM: h gains a 16-column pad whose first
+        # extra column is ones, so column H of dW_aug IS sum(dlog) — the
+        # separate db reduction re-read the whole dlog chunk (7.9 GB at
+        # the bench shape); 16 (not 1) keeps the GEMM N%16 alignment
+        Haug = H + 16 if has_bias else H
+        h_aug = h
+        if has_bias:
+            h_aug = torch.zeros(N, Haug, dtype=h.dtype, device=dev)
+            h_aug[:, :H] = h
+            h_aug[:, H] = 1
+        dw_aug = torch.zeros(V, Haug, dtype=torch.float32, device=dev)
         recompute = None if logits_full is not None else \
             torch.empty(min(C, N), V, dtype=h.dtype, device=dev)
         # (V, H) column-major e4m3 weight for the fp8 dh GEMM
@@ -180,19 +189,17 @@ class _FusedCEFp8Function(torch.autograd.Function):
             else:
                 torch.mm(dlog, weight, out=dh[s:e])
             if overlap:
-                # dW/db of THIS chunk overlap the next chunk's epilogue +
-                # dh (disjoint slices of the resident buffer)
+                # dW(+db) of THIS chunk overlaps the next chunk's
+                # epilogue + dh (disjoint slices of the resident buffer)
                 side.wait_stream(main)
                 with torch.cuda.stream(side):
-                    dw += torch.mm(dlog.t(), h[s:e])
-                    if has_bias:
-                        db += dlog.sum(dim=0).to(torch.float32)
+                    dw_aug += torch.mm(dlog.t(), h_aug[s:e])
             else:
-                dw += torch.mm(dlog.t(), h[s:e])
-                if has_bias:
-                    db += dlog.sum(dim=0).to(torch.float32)
+                dw_aug += torch.mm(dlog.t(), h_aug[s:e])
         if overlap:
             main.wait_stream(side)
+        dw = dw_aug[:, :H]
+        db = dw_aug[:, H] if has_bias else None
         return (dh, dw.to(weight.dtype),
                 db.to(ctx.bias_dtype) if has_bias else None, None)
 
@@ -261,9 +268,9 @@ class _FusedCEFunction(torch.autograd.Function):
             # in-place: dlog <- (softmax(dlog + bias) - onehot) * scale
             lib.ce_dlogits(dlog, targets[s:e], bias_arg, lse[s:e], scale)
             torch.mm(dlog, weight, out=dh[s:e])
-            dw += torch.mm(dlog.t(), h[s:e])
-            if has_bias:
-                db += dlog.sum(dim=0).to(torch.float32)
+            dw_aug += torch.mm(dlog.t(), h_aug[s:e])
+        dw = dw_aug[:, :H]
+        db = dw_aug[:, H] if has_bias else None
         return (dh, dw.to(weight.dtype),
                 db.to(ctx.bias_dtype) if has_bias else None, None)
 
