@@ -251,13 +251,20 @@ class _FusedCEFunction(torch.autograd.Function):
         V = weight.shape[0]
         C = _FusedCEFunction.CHUNK
         dh = torch.empty_like(h)
-        dw = torch.zeros_like(weight, dtype=torch.float32)
-        db = torch.zeros(V, dtype=torch.float32, device=h.device) if has_bias else None
         scale = (dloss / N).to(torch.float32).reshape(1)
         scratch = None if logits_full is not None else \
             torch.empty(min(C, N), V, dtype=h.dtype, device=h.device)
         w_t = weight.t()
         bias_arg = b32 if has_bias else _empty_f32(h.device)
+        # db folded into the dW GEMM via a ones-column (see the fp8
+        # variant above for the rationale and the 16-column alignment pad)
+        Haug = H + 16 if has_bias else H
+        h_aug = h
+        if has_bias:
+            h_aug = torch.zeros(N, Haug, dtype=h.dtype, device=h.device)
+            h_aug[:, :H] = h
+            h_aug[:, H] = 1
+        dw_aug = torch.zeros(V, Haug, dtype=torch.float32, device=h.device)
         for s in range(0, N, C):
             e = min(N, s + C)
             if logits_full is not None:
