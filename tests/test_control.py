@@ -195,3 +195,48 @@ def test_pipeline_runner_two_step_contract(tmp_path):
     rec2 = PipelineRunner(spec, runner=fail_first).run()
     assert rec2["status"] == "Failed"
     assert len(rec2["steps"]) == 1 and rec2["steps"][0]["returncode"] == 3
+
+
+def test_pipeline_cli_main(tmp_path):
+    """CLI entry: runs a pipeline YAML end-to-end as subprocesses and
+    exits nonzero on step failure."""
+    import json
+    import subprocess
+    import sys
+    from pathlib import Path
+    pipe = tmp_path / "p.yaml"
+    pipe.write_text("""
+name: demo
+params: {msg: world}
+steps:
+  - name: one
+    command: [python, -c, "print('hello {msg}')"]
+  - name: two
+    command: [python, -c, "import sys; sys.exit(0)"]
+""")
+    root = Path(__file__).resolve().parents[1]
+    r = subprocess.run(
+        [sys.executable, "-m", "code_intelligence_amd.control.pipeline",
+         str(pipe), "--run_dir", str(tmp_path / "runs"),
+         "--param", "msg=there"],
+        capture_output=True, text=True, timeout=120, cwd=root)
+    assert r.returncode == 0, r.stderr
+    out = json.loads(r.stdout.strip().splitlines()[-1])
+    assert out["status"] == "Succeeded"
+    rec = json.loads(next((tmp_path / "runs").glob("*.json")).read_text())
+    assert rec["params"]["msg"] == "there"
+
+    bad = tmp_path / "bad.yaml"
+    bad.write_text("""
+name: demo2
+steps:
+  - name: boom
+    command: [python, -c, "import sys; sys.exit(3)"]
+  - name: never
+    command: [python, -c, "print('x')"]
+""")
+    r2 = subprocess.run(
+        [sys.executable, "-m", "code_intelligence_amd.control.pipeline",
+         str(bad), "--run_dir", str(tmp_path / "runs2")],
+        capture_output=True, text=True, timeout=120, cwd=root)
+    assert r2.returncode == 1
