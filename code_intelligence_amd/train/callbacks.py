@@ -155,3 +155,33 @@ class TerminateOnNaN(Callback):
             raise FloatingPointError(
                 f"non-finite training loss {loss} at step {step} "
                 f"(lr={getattr(trainer, 'last_lr', None)})")
+
+
+class PeriodicCheckpoint(Callback):
+    """Full-state checkpoint every N optimizer steps (mid-epoch).
+
+    The reference only checkpointed per-epoch via SaveModelCallback; at
+    the reference's pretraining scale (a week of one-cycle on 16.7M
+    issues) an epoch is hours, so step-granular resume matters. Writes
+    ``<path>.tmp`` then atomically renames, keeps the last ``keep``
+    files (``step-<N>.ckpt``)."""
+
+    def __init__(self, out_dir, every_steps: int = 1000, keep: int = 2):
+        from pathlib import Path
+        self.out_dir = Path(out_dir)
+        self.every = max(1, int(every_steps))
+        self.keep = keep
+
+    def on_step_end(self, trainer, step: int, loss: float):
+        if step == 0 or step % self.every:
+            return
+        import os
+        self.out_dir.mkdir(parents=True, exist_ok=True)
+        path = self.out_dir / f"step-{step}.ckpt"
+        tmp = path.with_suffix(".ckpt.tmp")
+        trainer.save_checkpoint(tmp)
+        os.replace(tmp, path)
+        ckpts = sorted(self.out_dir.glob("step-*.ckpt"),
+                       key=lambda p: int(p.stem.split("-")[1]))
+        for old in ckpts[: -self.keep]:
+            old.unlink(missing_ok=True)

@@ -19,7 +19,7 @@ import torch
 from ..data.lm_loader import LMStreamLoader
 from ..data.synthetic import synthetic_issue_tokens
 from ..models.awd_lstm import AWDLSTM
-from .callbacks import (CSVLogger, EarlyStopping, JSONRunLogger,
+from .callbacks import (PeriodicCheckpoint, CSVLogger, EarlyStopping, JSONRunLogger,
                         ReduceLROnPlateau, SaveModel)
 from .trainer import LMTrainer, TrainConfig
 
@@ -48,6 +48,9 @@ def build_argparser() -> argparse.ArgumentParser:
                    help="checkpoint path (trainer.save_checkpoint) to resume "
                         "from: restores model+optimizer+epoch+RNG and "
                         "continues the same trajectory")
+    p.add_argument("--checkpoint_every", type=int, default=0,
+                   help="full-state checkpoint every N steps (mid-epoch "
+                        "resume for long pretraining runs); 0 = off")
     p.add_argument("--save_checkpoint", type=str, default=None,
                    help="write a full resume checkpoint here after training")
     return p
@@ -101,7 +104,8 @@ def main(argv=None) -> dict:
         ReduceLROnPlateau(patience=1),
         CSVLogger(out / "history.csv"),
         JSONRunLogger(out / "run.jsonl", config=vars(args)),
-    ])
+    ] + ([PeriodicCheckpoint(out / "checkpoints", args.checkpoint_every)]
+         if args.checkpoint_every else []))
     if args.resume:
         trainer.load_checkpoint(args.resume, map_location=device)
     metrics = trainer.fit(train_loader, valid_loader,
