@@ -79,7 +79,9 @@ def serve_bench(args, quick: bool = True) -> dict:
     from code_intelligence_amd.models.awd_lstm import AWDLSTM
     model = AWDLSTM(vocab_sz=len(v), emb_sz=emb, n_hid=hid, n_layers=layers,
                     qrnn=args.qrnn)
-    w = InferenceWrapper(encoder=model.encoder, vocab=v)
+    w = InferenceWrapper(encoder=model.encoder, vocab=v,
+                         use_graphs=os.environ.get("CI_SERVE_GRAPHS",
+                                                   "0") == "1")
     n_bulk, n_single = (1500, 30) if quick else (2000, 100)
     if not on_gpu:
         n_bulk, n_single = 40, 5
