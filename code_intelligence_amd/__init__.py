@@ -13,4 +13,4 @@ label-prediction microservices), designed MI355X-first:
 Layer map mirrors /root/repo/SURVEY.md §1; kernel inventory SURVEY.md §2.4.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
