@@ -166,9 +166,14 @@ class LMTrainer:
         self.global_step = ckpt.get("global_step", 0)
         self.epoch = ckpt.get("epoch", 0)
         if "rng" in ckpt:
-            torch.set_rng_state(ckpt["rng"])
+            # RNG states must stay CPU ByteTensors even when the rest of
+            # the checkpoint is mapped to cuda (map_location="cuda"
+            # otherwise breaks set_rng_state — caught by the GPU
+            # deployed-shape resume check, round 2)
+            torch.set_rng_state(ckpt["rng"].cpu().to(torch.uint8))
         if "rng_cuda" in ckpt and torch.cuda.is_available():
-            torch.cuda.set_rng_state_all(ckpt["rng_cuda"])
+            torch.cuda.set_rng_state_all(
+                [s.cpu().to(torch.uint8) for s in ckpt["rng_cuda"]])
 
     def fit(self, train_loader, valid_loader=None, epochs: int = 1,
             one_cycle: Optional[bool] = None) -> dict:

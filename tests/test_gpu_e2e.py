@@ -249,3 +249,42 @@ def test_qrnn_bulk_serve_shape_no_fault():
         h, cT = qrnn_forward(x, c0, w, b, window=1)
     torch.cuda.synchronize()
     assert h.shape == (B, T, H) and torch.isfinite(h.float().sum())
+
+
+@pytest.mark.timeout(300)
+def test_checkpoint_resume_with_cuda_map_location(tmp_path):
+    """load_checkpoint(map_location='cuda') must restore RNG states (they
+    are CPU ByteTensors; a cuda map_location used to break
+    torch.set_rng_state — caught in round 2 at the deployed shape)."""
+    import torch
+    from code_intelligence_amd.models.awd_lstm import AWDLSTM
+    from code_intelligence_amd.train.trainer import LMTrainer, TrainConfig
+
+    def build():
+        torch.manual_seed(0)
+        m = AWDLSTM(vocab_sz=500, emb_sz=32, n_hid=48, n_layers=2
+                    ).to("cuda", torch.bfloat16)
+        return m, LMTrainer(m, TrainConfig(one_cycle=False))
+
+    g = torch.Generator().manual_seed(3)
+
+    def step(t):
+        x = torch.randint(2, 500, (8, 12), generator=g).cuda()
+        return t.train_step(x, torch.roll(x, -1, 1), 1e-3)
+
+    m, tr = build()
+    m.train(); m.reset(8)
+    for _ in range(4):
+        step(tr)
+    tr.save_checkpoint(tmp_path / "c.ckpt")
+    a = [step(tr) for _ in range(2)]
+
+    m2, tr2 = build()
+    tr2.load_checkpoint(tmp_path / "c.ckpt", map_location="cuda")
+    assert tr2.global_step == 4
+    m2.train(); m2.reset(8)
+    g = torch.Generator().manual_seed(3)
+    for _ in range(4):
+        x = torch.randint(2, 500, (8, 12), generator=g)  # skip consumed
+    b = [step(tr2) for _ in range(2)]
+    assert all(abs(x - y) < 0.25 for x, y in zip(a, b)), (a, b)
