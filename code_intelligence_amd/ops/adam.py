@@ -23,6 +23,19 @@ class FusedAdamW(torch.optim.Optimizer):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
 
+    def load_state_dict(self, state_dict):
+        """torch's Optimizer.load_state_dict casts floating state to each
+        PARAM's dtype — for bf16 params that silently downcasts the fp32
+        master weights/moments on every resume (and the fused kernel then
+        rejects the bf16 masters). Restore fp32 after the base load.
+        Caught by the GPU deployed-shape resume check (round 2)."""
+        super().load_state_dict(state_dict)
+        for st in self.state.values():
+            for k in ("master", "exp_avg", "exp_avg_sq"):
+                v = st.get(k)
+                if torch.is_tensor(v) and v.dtype != torch.float32:
+                    st[k] = v.to(torch.float32)
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
